@@ -1,0 +1,316 @@
+"""
+Local file datasource: scan / build / query / index-scan / index-read.
+
+Orchestration mirrors the reference file backend
+(reference lib/datasource-file.js): enumerate files (time-format pruned),
+run the scan engine over their bytes, fan one parse pass into one
+aggregation per metric for index builds, route aggregated points into
+per-interval SQLite index files, and answer queries by merging per-file
+index partials through a final aggregator.
+
+The scan engine is pluggable: the CPU oracle (engine/cpu.py) or the HIP
+GPU engine (engine/gpu.py).
+"""
+
+import math
+import os
+import sys
+
+from .. import jsdate
+from .. import krill
+from ..fsfind import FindCounters, find_data_files
+from ..index import IndexQuerier, IndexSink
+from ..index.query import IndexError_
+from ..points import Aggregator
+from ..query import QueryConfig
+
+# interval -> (iso prefix length, suffix completing a full timestamp,
+#              subdir, strftime file pattern);
+# reference lib/datasource-file.js:466-476, lib/dragnet-impl.js:194-236
+INTERVALS = {
+    "hour": (len("2014-07-02T00"), ":00:00Z", "by_hour",
+             "%Y-%m-%d-%H.sqlite", 3600),
+    "day": (len("2014-07-02"), "T00:00:00Z", "by_day",
+            "%Y-%m-%d.sqlite", 86400),
+}
+
+
+def metric_query(metric, interval, time_field, after_ms=None,
+                 before_ms=None):
+    """Build the QueryConfig for one metric's index aggregation
+    (reference lib/dragnet-impl.js:290-323): for chunked intervals,
+    prepend a reserved __dn_ts lquantize breakdown over the time field.
+    """
+    breakdowns = [dict(b) for b in metric.get("breakdowns", [])]
+    if interval != "all":
+        step = INTERVALS[interval][4]
+        breakdowns.insert(0, {
+            "name": "__dn_ts", "aggr": "lquantize", "step": step,
+            "field": time_field, "date": "",
+        })
+    return QueryConfig(
+        filter=metric.get("filter"),
+        breakdowns=breakdowns,
+        time_after=_ms_to_iso(after_ms),
+        time_before=_ms_to_iso(before_ms),
+        allow_reserved=True)
+
+
+def _ms_to_iso(ms):
+    return None if ms is None else jsdate.to_iso(ms / 1000.0)
+
+
+class ScanResult(object):
+    """A scan's output: per-query aggregators + counter stages."""
+
+    def __init__(self, aggregators, stages, files=None):
+        self.aggregators = aggregators
+        self.stages = stages
+        self.files = files or []
+
+
+class FileDatasource(object):
+    def __init__(self, ds, engine=None):
+        self.ds = ds
+        self._engine = engine
+
+    def close(self):
+        pass
+
+    def engine(self):
+        if self._engine is None:
+            from ..engine import get_engine
+            self._engine = get_engine()
+        return self._engine
+
+    # ---- raw data scanning ----
+
+    def _find_data(self, query, counters=None, warn=None):
+        if (query.before_ms is not None
+                and self.ds.time_field is None):
+            raise ValueError(
+                'datasource is missing "timefield" for "before" and '
+                '"after" constraints')
+        if (query.before_ms is not None
+                and self.ds.time_format is None):
+            sys.stderr.write(
+                'warn: datasource is missing "timeformat" for '
+                '"before" and "after" constraints\n')
+        return find_data_files(
+            self.ds.path, self.ds.time_format,
+            query.after_ms, query.before_ms,
+            counters=counters, warn=warn)
+
+    def scan(self, query, dry_run=False, out=None):
+        """Run one query over the raw data.  Returns a ScanResult with a
+        single aggregator (or prints the file list for dry runs)."""
+        counters = FindCounters()
+        files = list(self._find_data(query, counters=counters))
+        if dry_run:
+            out = out or sys.stderr
+            out.write("would scan files:\n")
+            for path, _ in files:
+                out.write("    %s\n" % path)
+            return None
+        result = self.engine().scan(
+            files=[p for p, _ in files],
+            queries=[query],
+            ds_filter=self.ds.filter,
+            time_field=self.ds.time_field,
+            data_format=self.ds.data_format)
+        result.stages = counters.stages() + result.stages
+        result.files = [p for p, _ in files]
+        return result
+
+    def scan_multi(self, queries, after_ms, before_ms, dry_run=False,
+                   out=None):
+        """One parse pass, N metric aggregations (index builds;
+        reference lib/datasource-file.js:386-432)."""
+        pseudo = QueryConfig(time_after=_ms_to_iso(after_ms),
+                             time_before=_ms_to_iso(before_ms))
+        counters = FindCounters()
+        files = list(self._find_data(pseudo, counters=counters))
+        if dry_run:
+            out = out or sys.stderr
+            out.write("would scan files:\n")
+            for path, _ in files:
+                out.write("    %s\n" % path)
+            return None
+        result = self.engine().scan(
+            files=[p for p, _ in files],
+            queries=queries,
+            ds_filter=self.ds.filter,
+            time_field=self.ds.time_field,
+            data_format=self.ds.data_format)
+        result.stages = counters.stages() + result.stages
+        result.files = [p for p, _ in files]
+        return result
+
+    # ---- index building ----
+
+    def metric_queries(self, metrics, interval, after_ms, before_ms):
+        if interval not in ("all", "hour", "day"):
+            raise ValueError('unsupported interval: "%s"' % interval)
+        if interval != "all" and self.ds.time_field is None:
+            raise ValueError(
+                'datasource is missing "timefield" needed for '
+                'interval "%s"' % interval)
+        return [metric_query(m, interval, self.ds.time_field,
+                             after_ms, before_ms)
+                for m in metrics]
+
+    def index_scan_points(self, metrics, interval, after_ms=None,
+                          before_ms=None, dry_run=False):
+        """The distributed-build map phase: scan raw data, emit points
+        tagged with __dn_metric (and __dn_ts for chunked intervals).
+        Returns (points, stages) or None for dry runs."""
+        queries = self.metric_queries(metrics, interval, after_ms,
+                                      before_ms)
+        result = self.scan_multi(queries, after_ms, before_ms,
+                                 dry_run=dry_run)
+        if result is None:
+            return None
+        points = []
+        for qi, agg in enumerate(result.aggregators):
+            for p in agg.points():
+                p["fields"]["__dn_metric"] = qi
+                points.append(p)
+        return points, result.stages
+
+    def build(self, metrics, interval="day", after_ms=None,
+              before_ms=None, dry_run=False):
+        """Materialize indexes (reference lib/datasource-file.js:307-432).
+        Returns the list of index files written."""
+        if not self.ds.index_path:
+            raise ValueError(
+                'datasource is missing "indexPath" for index operations')
+        rv = self.index_scan_points(metrics, interval, after_ms,
+                                    before_ms, dry_run=dry_run)
+        if rv is None:
+            return None
+        points, _stages = rv
+        return write_index(self.ds.index_path, metrics, interval, points)
+
+    # ---- index querying ----
+
+    def _find_index_files(self, query, interval):
+        if interval == "all":
+            root = os.path.join(self.ds.index_path, "all")
+            return find_files([root])
+        _, _, subdir, pattern, _ = INTERVALS[interval]
+        root = os.path.join(self.ds.index_path, subdir)
+        return find_data_files(root, pattern, query.after_ms,
+                               query.before_ms)
+
+    def query(self, query, interval="day", dry_run=False, out=None):
+        """Answer a query from the index tree: per-file partials merged
+        through a final aggregator (reference
+        lib/datasource-file.js:573-691)."""
+        if not self.ds.index_path:
+            raise ValueError(
+                'datasource is missing "indexPath" for index operations')
+        files = list(self._find_index_files(query, interval))
+        if dry_run:
+            out = out or sys.stderr
+            out.write("would scan files:\n")
+            for path, _ in files:
+                out.write("    %s\n" % path)
+            return None
+
+        # The datasource filter is NOT applied at query time: it was
+        # already applied when the index was built (observed in
+        # tst.index_file.sh: `dn query` on a GET-filtered datasource
+        # returns the GET subset with no filter in the query).
+        eff = QueryConfig(
+            filter=query.filter,
+            breakdowns=[dict(b) for b in query.breakdowns],
+            time_after=_ms_to_iso(query.after_ms),
+            time_before=_ms_to_iso(query.before_ms),
+            allow_reserved=True)
+
+        final = Aggregator(query)
+        nerrors = []
+        for path, _st in files:
+            try:
+                iq = IndexQuerier(path)
+            except IndexError_ as e:
+                nerrors.append((path, str(e)))
+                continue
+            try:
+                partial = iq.run(eff)
+                for p in partial.points():
+                    final.write(p)
+            except IndexError_ as e:
+                nerrors.append((path, str(e)))
+            finally:
+                iq.close()
+        stages = [("IndexQuery", {
+            "nfiles": len(files), "nerrors": len(nerrors)})]
+        result = ScanResult([final], stages, [p for p, _ in files])
+        result.errors = nerrors
+        return result
+
+    def index_read_points(self, metrics, interval="day"):
+        """Emit every stored row of every index file as tagged points
+        (the distributed-build reduce input; also `dn index-read`)."""
+        pseudo = QueryConfig()
+        files = list(self._find_index_files(pseudo, interval))
+        for path, _st in files:
+            iq = IndexQuerier(path)
+            try:
+                for mi, met in enumerate(iq.metrics):
+                    tbl = "dragnet_index_%d" % met["id"]
+                    cols = [b["name"] for b in met["params"]]
+                    sql = "SELECT %s from %s" % (
+                        ", ".join(["%s" % _esc(c) for c in cols]
+                                  + ["value"]), tbl)
+                    for row in iq.db.execute(sql):
+                        fields = {}
+                        for c, v in zip(cols, row[:-1]):
+                            fields[c] = v
+                        fields["__dn_metric"] = mi
+                        yield {"fields": fields, "value": row[-1]}
+            finally:
+                iq.close()
+
+
+def _esc(c):
+    from ..index.sink import sqlite3_escape
+    return sqlite3_escape(c)
+
+
+def find_files(roots):
+    from ..fsfind import find_files as ff
+    return ff(roots)
+
+
+def write_index(index_path, metrics, interval, points):
+    """Route aggregated points into per-interval IndexSinks; atomic
+    rename on flush.  Returns index file paths written."""
+    written = []
+    if interval == "all":
+        sink = IndexSink(os.path.join(index_path, "all"), metrics)
+        for p in points:
+            sink.write_point(p)
+        sink.flush()
+        return [sink.filename]
+
+    prefixlen, suffix, subdir, _pat, _step = INTERVALS[interval]
+    root = os.path.join(index_path, subdir)
+    sinks = {}
+    for p in points:
+        dnts = p["fields"]["__dn_ts"]
+        assert isinstance(dnts, (int, float)) and not math.isnan(dnts)
+        datestr = jsdate.to_iso(dnts)
+        bucketname = datestr[:prefixlen]
+        if bucketname not in sinks:
+            label = bucketname.replace("T", "-")
+            start = jsdate.parse_ms(bucketname + suffix) // 1000
+            sinks[bucketname] = IndexSink(
+                os.path.join(root, label + ".sqlite"), metrics,
+                config={"dn_start": start})
+        sinks[bucketname].write_point(p)
+    for name in sorted(sinks):
+        sinks[name].flush()
+        written.append(sinks[name].filename)
+    return written

@@ -1,0 +1,167 @@
+"""
+Skinner-style streaming aggregation: points, the Aggregator, the Flattener.
+
+A "point" is {'fields': {...}, 'value': n}.  Aggregation of points is
+associative and commutative (value-weighted re-aggregation), which is the
+load-bearing semantic that makes both the index subsystem and the multi-GPU
+RCCL merge correct (reference: the x3 idempotence test
+tests/dn/local/tst.format_skinner.sh:25-37 and the Manta reduce phase
+lib/datasource-manta.js:212-219).
+
+Canonical grouping values per breakdown column (derived from reference
+golden outputs, e.g. tests/dn/local/tst.scan_file.sh.out:136-147 where
+null/missing req.caller group as the strings "null"/"undefined"):
+
+  * aggregated (quantize/lquantize) columns -> ordinal bucket index (int);
+    emitted in points as the bucket minimum (a number)
+  * date (synthetic, non-aggregated) columns -> unix seconds (int)
+  * all other columns -> strings; numbers/booleans/null/missing coerce via
+    JavaScript string conversion ("200", "true", "null", "undefined")
+"""
+
+from .krill import MISSING, pluck
+
+
+def js_num_str(v):
+    """JavaScript String(number) for the values we encounter."""
+    if isinstance(v, bool):
+        return "true" if v else "false"
+    if isinstance(v, int):
+        return str(v)
+    if isinstance(v, float):
+        if v != v:
+            return "NaN"
+        if v == float("inf"):
+            return "Infinity"
+        if v == float("-inf"):
+            return "-Infinity"
+        if v == int(v) and abs(v) < 2 ** 53:
+            return str(int(v))
+        return repr(v)
+    return str(v)
+
+
+def lookup(fields, name):
+    """Field lookup for aggregation: literal key first, then dotted path.
+
+    The literal-first order is what lets skinner-format points (whose
+    fields are literally keyed "req.method") re-aggregate identically to
+    raw nested records.
+    """
+    if isinstance(fields, dict) and name in fields:
+        return fields[name]
+    return pluck(fields, name)
+
+
+def canonical(val, has_date=False):
+    """Canonical grouping value for a NON-aggregated column."""
+    if val is MISSING:
+        return "undefined"
+    if val is None:
+        return "null"
+    if isinstance(val, bool):
+        return "true" if val else "false"
+    if has_date:
+        # synthetic date fields carry unix seconds; keep numeric
+        return int(val) if isinstance(val, (int, float)) else val
+    if isinstance(val, (int, float)):
+        return js_num_str(val)
+    if isinstance(val, str):
+        return val
+    # objects/arrays as group-by values: JS coerces to
+    # "[object Object]" / join(',') — model the common cases
+    if isinstance(val, dict):
+        return "[object Object]"
+    if isinstance(val, list):
+        return ",".join(canonical(x) for x in val)
+    return str(val)
+
+
+class Aggregator(object):
+    """Streaming group-by/count over points.
+
+    Internal keys hold ordinal bucket indices for aggregated columns.
+    Records whose aggregated column is non-numeric are dropped
+    ('nonnumeric' counter; reference README.md:718-722).
+    """
+
+    def __init__(self, query):
+        self.query = query
+        self.breakdowns = query.breakdowns
+        self.bucketizers = query.bucketizers
+        self.table = {}
+        self.ninputs = 0
+        self.ndropped_nonnumeric = 0
+
+    def write(self, point):
+        self.ninputs += 1
+        fields = point["fields"]
+        key = []
+        for b in self.breakdowns:
+            name = b["name"]
+            val = lookup(fields, name)
+            bk = self.bucketizers.get(name)
+            if bk is not None:
+                if isinstance(val, bool) or not isinstance(val, (int, float)):
+                    self.ndropped_nonnumeric += 1
+                    return False
+                key.append(bk.bucket(val))
+            else:
+                key.append(canonical(val, "date" in b))
+        k = tuple(key)
+        self.table[k] = self.table.get(k, 0) + point["value"]
+        return True
+
+    def noutputs(self):
+        # A zero-breakdown aggregation always emits exactly one point,
+        # even over empty input (observed tst.empty.sh.out:1-24).
+        if not self.breakdowns:
+            return 1
+        return len(self.table)
+
+    def points(self):
+        """Emit aggregated results as points (sorted for determinism).
+
+        Aggregated fields carry bucket minimums (numbers), matching the
+        reference's resultsAsPoints output
+        (tests/dn/local/tst.scan_file.sh.out:306-314).
+        """
+        if not self.breakdowns:
+            total = sum(self.table.values()) if self.table else 0
+            return [{"fields": {}, "value": total}]
+        out = []
+        for key in sorted(self.table.keys(), key=_sort_key):
+            fields = {}
+            for b, kv in zip(self.breakdowns, key):
+                bk = self.bucketizers.get(b["name"])
+                fields[b["name"]] = bk.bucket_min(kv) if bk else kv
+            out.append({"fields": fields, "value": self.table[key]})
+        return out
+
+    def rows(self):
+        """Flattened rows: tuples + value, aggregated columns as ORDINAL
+        bucket indices (expanded at print time).  Zero breakdowns ->
+        [total] (a bare number), mirroring the reference flattener."""
+        if not self.breakdowns:
+            total = sum(self.table.values()) if self.table else 0
+            return [total]
+        return [list(k) + [v] for k, v in self.table.items()]
+
+
+def _sort_key(key_tuple):
+    return tuple((0, v) if isinstance(v, (int, float)) else (1, v)
+                 for v in key_tuple)
+
+
+class Flattener(object):
+    """Re-aggregates a point stream into one flattened `rows` payload
+    (reference lib/skinner-flattener.js:20-35)."""
+
+    def __init__(self, query):
+        self.agg = Aggregator(query)
+
+    def write(self, point):
+        return self.agg.write(point)
+
+    def rows(self):
+        return self.agg.rows()

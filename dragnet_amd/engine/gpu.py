@@ -1,0 +1,359 @@
+"""
+MI355X GPU scan engine — host orchestration.
+
+Streams NDJSON bytes through the fused CDNA4 scan kernel
+(ops/hip/scan_kernels.hip): chunked reads into pinned host buffers,
+async H2D on the current HIP stream, torch-side newline indexing, one
+fused kernel launch per chunk, then table/dictionary extraction and
+host-side decode into the same canonical aggregate representation the
+CPU oracle produces (points.Aggregator) — so every downstream consumer
+(formatters, index sink, RCCL merge) is engine-agnostic.
+
+Capacity handling: the hash tables and dictionaries are sized up front;
+on overflow (C_OVERFLOW counter) the whole scan is retried with 8x the
+capacity (aggregation is a pure function of the input, so a restart is
+correct).
+"""
+
+import json
+import os
+
+import numpy as np
+
+from ..points import Aggregator, js_num_str
+from . import plan as planmod
+
+COUNTER_NAMES = ["lines", "invalid_json", "parsed", "ds_filtered",
+                 "ds_failedeval", "overflow"]
+C_GLOBAL_N = 8
+CM_N = 8
+(CM_FILTER_IN, CM_FILTERED, CM_FAILEDEVAL, CM_UNDEF, CM_BADDATE,
+ CM_TIME_OUT, CM_AGG_IN, CM_NONNUMERIC) = range(8)
+
+MAX_KEY = 8
+
+
+def _env_int(name, default):
+    try:
+        return int(os.environ.get(name, default))
+    except ValueError:
+        return default
+
+
+class GpuEngine(object):
+    name = "gpu"
+
+    def __init__(self, device=None):
+        import torch
+        self.torch = torch
+        if not torch.cuda.is_available():
+            raise RuntimeError("no GPU visible (torch.cuda unavailable)")
+        from ..ops import load_ops
+        self.ops = load_ops(required=True)
+        self.device = device if device is not None else \
+            torch.device("cuda", torch.cuda.current_device())
+        self.chunk_bytes = _env_int("DRAGNET_CHUNK_MB", 64) * 1024 * 1024
+
+    # ---- public engine interface ----
+
+    def scan(self, files, queries, ds_filter=None, time_field=None,
+             data_format="json", byte_source=None):
+        from ..datasource.file import ScanResult
+
+        cplan = planmod.compile_plan(
+            queries, ds_filter=ds_filter, time_field=time_field,
+            data_format=data_format)
+
+        agg_slots = _env_int("DRAGNET_AGG_SLOTS", 1 << 20)
+        dict_slots = _env_int("DRAGNET_DICT_SLOTS", 1 << 20)
+        dict_data = _env_int("DRAGNET_DICT_DATA_MB", 256) * 1024 * 1024
+
+        for attempt in range(3):
+            ctx = _ScanContext(self, cplan, agg_slots, dict_slots,
+                               dict_data)
+            src = byte_source if byte_source is not None \
+                else _read_files(files, self.chunk_bytes)
+            if byte_source is not None and attempt > 0:
+                raise RuntimeError(
+                    "table overflow on non-restartable byte source")
+            for chunk in src:
+                ctx.feed(chunk)
+            ctx.flush()
+            if ctx.overflowed():
+                agg_slots *= 8
+                dict_slots *= 8
+                dict_data *= 4
+                continue
+            aggs, stages = ctx.finalize(queries)
+            return ScanResult(aggs, stages)
+        raise RuntimeError("aggregation tables overflowed after retries")
+
+
+class _ScanContext(object):
+    def __init__(self, eng, cplan, agg_slots, dict_slots, dict_data_cap):
+        torch = eng.torch
+        self.eng = eng
+        self.t = torch
+        self.cplan = cplan
+        dev = eng.device
+        i32 = dict(dtype=torch.int32, device=dev)
+        f64 = dict(dtype=torch.float64, device=dev)
+        u8 = dict(dtype=torch.uint8, device=dev)
+
+        def dev_i32(np_arr):
+            return torch.from_numpy(
+                np.ascontiguousarray(np_arr.astype(np.int32))).to(dev)
+
+        def dev_f64(np_arr):
+            return torch.from_numpy(
+                np.ascontiguousarray(np_arr.astype(np.float64))).to(dev)
+
+        # plan buffers
+        self.field_sigs = torch.from_numpy(
+            cplan.field_sigs.view(np.int64)).to(dev)
+        progs, bounds = cplan.programs
+        self.prog_nodes = dev_i32(progs)
+        self.prog_bounds = dev_i32(bounds)
+        self.const_meta = dev_i32(cplan.const_meta)
+        self.const_dvals = dev_f64(cplan.const_dvals)
+        self.const_bytes = torch.from_numpy(
+            np.ascontiguousarray(cplan.const_bytes)).to(dev)
+        self.synth_slots = dev_i32(cplan.synthetic)
+        metrics, sreq = cplan.metrics
+        self.metric_rows = dev_i32(metrics)
+        self.synth_req = dev_i32(sreq)
+        bds, steps = cplan.breakdown_descs
+        self.bd_rows = dev_i32(bds)
+        self.bd_steps = dev_f64(steps)
+        self.nm = metrics.shape[0]
+
+        # tables
+        self.tables = []
+        for m in range(self.nm):
+            state = torch.zeros(agg_slots, **i32)
+            keys = torch.zeros(agg_slots * MAX_KEY, **i32)
+            count = torch.zeros(agg_slots, **f64)
+            self.tables.append((state, keys, count))
+        descs = eng.ops.agg_descs_host(
+            [t[0] for t in self.tables], [t[1] for t in self.tables],
+            [t[2] for t in self.tables])
+        self.table_descs = descs.to(dev)
+
+        # dictionaries
+        self.sd = dict(
+            state=torch.zeros(dict_slots, **i32),
+            hash=torch.zeros(dict_slots, dtype=torch.int64, device=dev),
+            id=torch.zeros(dict_slots, **i32),
+            off=torch.zeros(dict_slots, **i32),
+            len=torch.zeros(dict_slots, **i32),
+            data=torch.zeros(dict_data_cap, **u8),
+            used=torch.zeros(1, **i32),
+            next=torch.zeros(1, **i32),
+        )
+        self.nd = dict(
+            state=torch.zeros(dict_slots, **i32),
+            bits=torch.zeros(dict_slots, dtype=torch.int64, device=dev),
+            id=torch.zeros(dict_slots, **i32),
+            next=torch.zeros(1, **i32),
+        )
+        self.counters = torch.zeros(
+            C_GLOBAL_N + self.nm * CM_N, dtype=torch.int64, device=dev)
+
+        self._partial = b""
+        self._pinned = None
+        self.agg_slots = agg_slots
+
+    # ---- chunk feeding ----
+
+    def feed(self, chunk):
+        """Consume a byte chunk (records split on newlines; partial
+        trailing line carried to the next chunk)."""
+        data = self._partial + chunk
+        cut = data.rfind(b"\n")
+        if cut < 0:
+            self._partial = data
+            return
+        self._partial = data[cut + 1:]
+        self._run(data[:cut + 1])
+
+    def flush(self):
+        if self._partial:
+            self._run(self._partial + b"\n")
+            self._partial = b""
+
+    def _run(self, buf):
+        torch = self.t
+        n = len(buf)
+        if n == 0:
+            return
+        padded = (n + 15) & ~15
+        if (self._pinned is None or self._pinned.numel() < padded):
+            cap = max(padded, self.eng.chunk_bytes + (1 << 20))
+            self._pinned = torch.empty(
+                cap, dtype=torch.uint8, pin_memory=True)
+        pin = self._pinned
+        pin[:n] = torch.frombuffer(bytearray(buf), dtype=torch.uint8)
+        pin[n:padded] = 10  # newline padding for the 16B cursor window
+
+        dev_data = pin[:padded].to(self.eng.device, non_blocking=True)
+        nl = (dev_data[:n] == 10).nonzero().flatten().to(torch.int32)
+        if nl.numel() == 0:
+            return
+        starts = torch.cat([
+            torch.zeros(1, dtype=torch.int32, device=nl.device),
+            nl[:-1] + 1])
+        ends = nl
+
+        self.eng.ops.scan_chunk(
+            dev_data, starts, ends,
+            self.field_sigs, self.prog_nodes, self.prog_bounds,
+            self.const_meta, self.const_dvals, self.const_bytes,
+            self.synth_slots, self.cplan.n_synth,
+            self.metric_rows, self.synth_req,
+            self.bd_rows, self.bd_steps,
+            self.cplan.value_slot,
+            getattr(self.cplan, "fields_slot", -1),
+            self.cplan.data_format == "json-skinner",
+            self.table_descs,
+            self.sd["state"], self.sd["hash"], self.sd["id"],
+            self.sd["off"], self.sd["len"], self.sd["data"],
+            self.sd["used"], self.sd["next"],
+            self.nd["state"], self.nd["bits"], self.nd["id"],
+            self.nd["next"],
+            self.counters)
+
+    def overflowed(self):
+        return int(self.counters[5].item()) > 0
+
+    # ---- results ----
+
+    def finalize(self, queries):
+        torch = self.t
+        torch.cuda.synchronize(self.eng.device)
+        cnt = self.counters.cpu().numpy()
+
+        # dictionaries
+        n_str = int(self.sd["next"].item())
+        n_num = int(self.nd["next"].item())
+        strings = []
+        if n_str:
+            off, ln = self.eng.ops.extract_strdict(
+                self.sd["state"], self.sd["hash"], self.sd["id"],
+                self.sd["off"], self.sd["len"], self.sd["data"],
+                self.sd["used"], self.sd["next"], n_str)
+            off = off.cpu().numpy().astype(np.uint32)
+            ln = ln.cpu().numpy().astype(np.uint32)
+            used = int(self.sd["used"].item())
+            blob = self.sd["data"][:used].cpu().numpy().tobytes()
+            for i in range(n_str):
+                raw = blob[off[i]:off[i] + ln[i]]
+                strings.append(_decode_json_string(raw))
+        numbers = np.zeros(0)
+        if n_num:
+            numbers = self.eng.ops.extract_numdict(
+                self.nd["state"], self.nd["bits"], self.nd["id"],
+                self.nd["next"], n_num).cpu().numpy()
+
+        aggs = []
+        for m, q in enumerate(queries):
+            state, keys, count = self.tables[m]
+            k, c = self.eng.ops.extract_agg(
+                state, keys, count, self.agg_slots)
+            k = k.cpu().numpy().astype(np.uint32)
+            c = c.cpu().numpy()
+            agg = Aggregator(q)
+            mc = cnt[C_GLOBAL_N + m * CM_N:
+                     C_GLOBAL_N + (m + 1) * CM_N]
+            agg.ninputs = int(mc[CM_AGG_IN])
+            agg.ndropped_nonnumeric = int(mc[CM_NONNUMERIC])
+            nk = len(q.breakdowns)
+            for i in range(k.shape[0]):
+                key = planmod.decode_key(k[i, :nk], q, strings, numbers)
+                v = c[i]
+                v = int(v) if float(v).is_integer() else float(v)
+                agg.table[key] = agg.table.get(key, 0) + v
+            aggs.append(agg)
+
+        stages = self._counter_stages(cnt, queries)
+        for name, c in stages:
+            if name == "Aggregator":
+                c["noutputs"] = aggs[0].noutputs()
+        return aggs, stages
+
+    def _counter_stages(self, cnt, queries):
+        """Reconstruct the reference pipeline counter stages from the
+        device counters (mirrors scan_cpu.ScanPipeline.counter_stages)."""
+        cp = self.cplan
+        stages = [("json parser", {
+            "ninputs": int(cnt[0]), "noutputs": int(cnt[2]),
+            "invalid json": int(cnt[1])})]
+        if cp.data_format == "json":
+            stages.append(("SkinnerAdapterStream",
+                           {"ninputs": int(cnt[2]),
+                            "noutputs": int(cnt[2])}))
+        # program 0 is the ds filter; OP_TRUE (8) means "no filter"
+        progs, bounds = cp.programs
+        if progs[bounds[0][0]][0] != 8:
+            n_in = int(cnt[2])
+            stages.append(("Datasource filter", {
+                "ninputs": n_in,
+                "noutputs": n_in - int(cnt[3]) - int(cnt[4]),
+                "nfilteredout": int(cnt[3]),
+                "nfailedeval": int(cnt[4])}))
+        # metric 0's pipeline (what the CLI displays for scans)
+        q = queries[0]
+        mc = cnt[C_GLOBAL_N:C_GLOBAL_N + CM_N]
+        n = int(mc[CM_FILTER_IN])
+        if q.filter is not None:
+            out = n - int(mc[CM_FILTERED]) - int(mc[CM_FAILEDEVAL])
+            stages.append(("User filter", {
+                "ninputs": n, "noutputs": out,
+                "nfilteredout": int(mc[CM_FILTERED]),
+                "nfailedeval": int(mc[CM_FAILEDEVAL])}))
+            n = out
+        metrics, _sreq = cp.metrics
+        if metrics[0][3] > 0:  # has synthetic requirements
+            out = n - int(mc[CM_UNDEF]) - int(mc[CM_BADDATE])
+            stages.append(("Datetime parser", {
+                "ninputs": n, "noutputs": out,
+                "undef": int(mc[CM_UNDEF]),
+                "baddate": int(mc[CM_BADDATE])}))
+            n = out
+        if metrics[0][5]:  # time filter
+            out = n - int(mc[CM_TIME_OUT])
+            stages.append(("Time filter", {
+                "ninputs": n, "noutputs": out,
+                "nfilteredout": int(mc[CM_TIME_OUT]),
+                "nfailedeval": 0}))
+            n = out
+        self._agg_stage_n = n
+        stages.append(("Aggregator", {
+            "ninputs": int(mc[CM_AGG_IN]),
+            "noutputs": 0,  # patched by caller if needed
+            "nonnumeric": int(mc[CM_NONNUMERIC])}))
+        return stages
+
+
+def _decode_json_string(raw):
+    """Decode raw in-record string bytes (may contain JSON escapes)
+    to the same Python string json.loads would produce."""
+    try:
+        s = raw.decode("utf-8")
+    except UnicodeDecodeError:
+        return raw.decode("utf-8", "replace")
+    if "\\" in s:
+        try:
+            return json.loads('"' + s + '"')
+        except ValueError:
+            return s
+    return s
+
+
+def _read_files(files, chunk_size):
+    for path in files:
+        with open(path, "rb", buffering=0) as f:
+            while True:
+                chunk = f.read(chunk_size)
+                if not chunk:
+                    break
+                yield chunk

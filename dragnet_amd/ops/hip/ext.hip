@@ -1,0 +1,227 @@
+// dragnet_amd MI355X scan engine — torch extension entry point.
+//
+// Single translation unit: includes the gfx950 kernels and exposes the
+// host launchers to Python.  Stateless C++: all device state (tables,
+// dictionaries, counters, plan buffers) lives in torch tensors owned by
+// the Python engine (dragnet_amd/engine/gpu.py).
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include "scan_kernels.hip"
+
+namespace dn {
+
+namespace {
+
+#define CHECK_GPU(t) TORCH_CHECK((t).is_cuda(), #t " must be on GPU")
+
+StrDict make_sdict(const torch::Tensor& state, const torch::Tensor& hash,
+                   const torch::Tensor& id, const torch::Tensor& off,
+                   const torch::Tensor& len, const torch::Tensor& data,
+                   const torch::Tensor& data_used,
+                   const torch::Tensor& next_id) {
+  StrDict d;
+  d.state = (uint32_t*)state.data_ptr();
+  d.hash = (uint64_t*)hash.data_ptr();
+  d.id = (uint32_t*)id.data_ptr();
+  d.off = (uint32_t*)off.data_ptr();
+  d.len = (uint32_t*)len.data_ptr();
+  d.nslots = (uint32_t)state.numel();
+  d.data = (uint8_t*)data.data_ptr();
+  d.data_cap = (uint32_t)data.numel();
+  d.data_used = (uint32_t*)data_used.data_ptr();
+  d.next_id = (uint32_t*)next_id.data_ptr();
+  return d;
+}
+
+NumDict make_ndict(const torch::Tensor& state, const torch::Tensor& bits,
+                   const torch::Tensor& id, const torch::Tensor& next_id) {
+  NumDict d;
+  d.state = (uint32_t*)state.data_ptr();
+  d.bits = (uint64_t*)bits.data_ptr();
+  d.id = (uint32_t*)id.data_ptr();
+  d.nslots = (uint32_t)state.numel();
+  d.next_id = (uint32_t*)next_id.data_ptr();
+  return d;
+}
+
+AggTable make_table(const torch::Tensor& state, const torch::Tensor& keys,
+                    const torch::Tensor& count) {
+  AggTable t;
+  t.state = (uint32_t*)state.data_ptr();
+  t.keys = (uint32_t*)keys.data_ptr();
+  t.count = (double*)count.data_ptr();
+  t.nslots = (uint32_t)state.numel();
+  return t;
+}
+
+hipStream_t current_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+}  // namespace
+
+void scan_chunk(
+    torch::Tensor data, torch::Tensor line_starts, torch::Tensor line_ends,
+    torch::Tensor field_sigs, torch::Tensor prog_nodes,
+    torch::Tensor prog_bounds, torch::Tensor const_meta,
+    torch::Tensor const_dvals, torch::Tensor const_bytes,
+    torch::Tensor synth_slots, int64_t n_synth,
+    torch::Tensor metric_rows, torch::Tensor synth_req,
+    torch::Tensor bd_rows, torch::Tensor bd_steps,
+    int64_t value_slot, int64_t fields_slot, bool skinner,
+    torch::Tensor table_descs,
+    torch::Tensor sd_state, torch::Tensor sd_hash, torch::Tensor sd_id,
+    torch::Tensor sd_off, torch::Tensor sd_len, torch::Tensor sd_data,
+    torch::Tensor sd_used, torch::Tensor sd_next,
+    torch::Tensor nd_state, torch::Tensor nd_bits, torch::Tensor nd_id,
+    torch::Tensor nd_next,
+    torch::Tensor counters) {
+  CHECK_GPU(data);
+  CHECK_GPU(line_starts);
+  CHECK_GPU(table_descs);
+  TORCH_CHECK(line_starts.numel() == line_ends.numel());
+  uint32_t nlines = (uint32_t)line_starts.numel();
+  if (nlines == 0) return;
+
+  ScanArgs A;
+  A.data = (const uint8_t*)data.data_ptr();
+  A.line_starts = (const uint32_t*)line_starts.data_ptr();
+  A.line_ends = (const uint32_t*)line_ends.data_ptr();
+  A.nlines = nlines;
+
+  A.P.field_sigs = (const uint64_t*)field_sigs.data_ptr();
+  A.P.nf = (int)field_sigs.numel();
+  A.P.prog_nodes = (const int32_t*)prog_nodes.data_ptr();
+  A.P.prog_bounds = (const int32_t*)prog_bounds.data_ptr();
+  A.P.const_meta = (const int32_t*)const_meta.data_ptr();
+  A.P.const_dvals = (const double*)const_dvals.data_ptr();
+  A.P.const_bytes = (const uint8_t*)const_bytes.data_ptr();
+  A.P.synth_slots = (const int32_t*)synth_slots.data_ptr();
+  A.P.ns = (int)n_synth;
+  A.P.metric_rows = (const int32_t*)metric_rows.data_ptr();
+  A.P.synth_req = (const int32_t*)synth_req.data_ptr();
+  A.P.bd_rows = (const int32_t*)bd_rows.data_ptr();
+  A.P.bd_steps = (const double*)bd_steps.data_ptr();
+  A.P.nm = (int)(metric_rows.numel() / 8);
+  A.P.value_slot = (int)value_slot;
+  A.P.fields_slot = (int)fields_slot;
+
+  A.tables = (AggTable*)table_descs.data_ptr();
+  A.sdict = make_sdict(sd_state, sd_hash, sd_id, sd_off, sd_len,
+                       sd_data, sd_used, sd_next);
+  A.ndict = make_ndict(nd_state, nd_bits, nd_id, nd_next);
+  A.counters = (unsigned long long*)counters.data_ptr();
+  A.data_format_skinner = skinner ? 1 : 0;
+
+  int nf = A.P.nf;
+  size_t lds = (size_t)nf * BLOCK * (8 + 4 + 4 + 1);
+  lds = (lds + 15) & ~(size_t)15;
+  lds += (size_t)LDS_CACHE * sizeof(LdsCacheEntry);
+  lds += (C_GLOBAL_N + (size_t)A.P.nm * CM_N) * 8;
+  lds += 64;  // slack
+  TORCH_CHECK(lds <= 160 * 1024, "plan needs too much LDS: ", lds);
+
+  uint32_t blocks = (nlines + BLOCK - 1) / BLOCK;
+  if (blocks > 4096) blocks = 4096;  // grid-stride the rest
+
+  hipLaunchKernelGGL(scan_kernel, dim3(blocks), dim3(BLOCK), lds,
+                     current_stream(), A);
+  hipError_t err = hipGetLastError();
+  TORCH_CHECK(err == hipSuccess, "scan_kernel launch failed: ",
+              hipGetErrorString(err));
+}
+
+// Build the device-side AggTable descriptor array from per-metric
+// tensors.  Returns a CPU byte tensor; the caller copies it to the GPU.
+torch::Tensor agg_descs_host(std::vector<torch::Tensor> states,
+                             std::vector<torch::Tensor> keys,
+                             std::vector<torch::Tensor> counts) {
+  int nm = (int)states.size();
+  auto out = torch::empty({(long)(nm * sizeof(AggTable))},
+                          torch::dtype(torch::kUInt8));
+  AggTable* descs = (AggTable*)out.data_ptr();
+  for (int m = 0; m < nm; m++)
+    descs[m] = make_table(states[m], keys[m], counts[m]);
+  return out;
+}
+
+// Compact one metric's table: returns (keys int32 [n, MAX_KEY],
+// counts f64 [n]).
+std::vector<torch::Tensor> extract_agg(torch::Tensor state,
+                                       torch::Tensor keys,
+                                       torch::Tensor count,
+                                       int64_t max_out) {
+  CHECK_GPU(state);
+  AggTable T = make_table(state, keys, count);
+  auto i32 = torch::dtype(torch::kInt32).device(state.device());
+  auto f64 = torch::dtype(torch::kFloat64).device(state.device());
+  auto out_keys = torch::zeros({max_out, (long)MAX_KEY}, i32);
+  auto out_counts = torch::zeros({max_out}, f64);
+  auto out_n = torch::zeros({1}, i32);
+  uint32_t blocks = (T.nslots + 255) / 256;
+  hipLaunchKernelGGL(extract_agg_kernel, dim3(blocks), dim3(256), 0,
+                     current_stream(), T, (uint32_t*)out_keys.data_ptr(),
+                     (double*)out_counts.data_ptr(),
+                     (uint32_t*)out_n.data_ptr(), (uint32_t)max_out);
+  hipError_t err = hipGetLastError();
+  TORCH_CHECK(err == hipSuccess, "extract_agg launch failed: ",
+              hipGetErrorString(err));
+  int64_t n = out_n.cpu().item<int32_t>();
+  if (n > max_out) n = max_out;
+  return {out_keys.narrow(0, 0, n), out_counts.narrow(0, 0, n)};
+}
+
+std::vector<torch::Tensor> extract_strdict(
+    torch::Tensor sd_state, torch::Tensor sd_hash, torch::Tensor sd_id,
+    torch::Tensor sd_off, torch::Tensor sd_len, torch::Tensor sd_data,
+    torch::Tensor sd_used, torch::Tensor sd_next, int64_t n_ids) {
+  StrDict D = make_sdict(sd_state, sd_hash, sd_id, sd_off, sd_len,
+                         sd_data, sd_used, sd_next);
+  auto i32 = torch::dtype(torch::kInt32).device(sd_state.device());
+  auto out_off = torch::zeros({n_ids > 0 ? n_ids : 1}, i32);
+  auto out_len = torch::zeros({n_ids > 0 ? n_ids : 1}, i32);
+  if (n_ids > 0) {
+    uint32_t blocks = (D.nslots + 255) / 256;
+    hipLaunchKernelGGL(extract_strdict_kernel, dim3(blocks), dim3(256), 0,
+                       current_stream(), D, (uint32_t*)out_off.data_ptr(),
+                       (uint32_t*)out_len.data_ptr(), (uint32_t)n_ids);
+    hipError_t err = hipGetLastError();
+    TORCH_CHECK(err == hipSuccess, "extract_strdict launch failed: ",
+                hipGetErrorString(err));
+  }
+  return {out_off, out_len};
+}
+
+torch::Tensor extract_numdict(torch::Tensor nd_state,
+                              torch::Tensor nd_bits, torch::Tensor nd_id,
+                              torch::Tensor nd_next, int64_t n_ids) {
+  NumDict D = make_ndict(nd_state, nd_bits, nd_id, nd_next);
+  auto out = torch::zeros(
+      {n_ids > 0 ? n_ids : 1},
+      torch::dtype(torch::kFloat64).device(nd_state.device()));
+  if (n_ids > 0) {
+    uint32_t blocks = (D.nslots + 255) / 256;
+    hipLaunchKernelGGL(extract_numdict_kernel, dim3(blocks), dim3(256), 0,
+                       current_stream(), D, (double*)out.data_ptr(),
+                       (uint32_t)n_ids);
+    hipError_t err = hipGetLastError();
+    TORCH_CHECK(err == hipSuccess, "extract_numdict launch failed: ",
+                hipGetErrorString(err));
+  }
+  return out;
+}
+
+}  // namespace dn
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "dragnet_amd MI355X scan engine (gfx950 HIP kernels)";
+  m.def("scan_chunk", &dn::scan_chunk, "fused NDJSON scan over one chunk");
+  m.def("agg_descs_host", &dn::agg_descs_host);
+  m.def("extract_agg", &dn::extract_agg);
+  m.def("extract_strdict", &dn::extract_strdict);
+  m.def("extract_numdict", &dn::extract_numdict);
+  m.attr("MAX_KEY") = (int)dn::MAX_KEY;
+  m.attr("MAX_FIELDS") = (int)dn::MAX_FIELDS;
+}

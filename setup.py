@@ -1,0 +1,32 @@
+"""
+Build the dragnet_amd gfx950 HIP extension in-tree:
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+Produces dragnet_amd/ops/_dragnet_hip.*.so (git-ignored; travels with
+the working tree to GPU machines).
+"""
+
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+setup(
+    name="dragnet_amd_ops",
+    version="0.1.0",
+    ext_modules=[
+        CUDAExtension(
+            name="dragnet_amd.ops._dragnet_hip",
+            sources=["dragnet_amd/ops/hip/ext.hip"],
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension},
+)

@@ -15,10 +15,10 @@ def test_constants_match_header():
     src = open(HDR).read()
 
     def cval(name):
-        m = re.search(r"%s\s*=\s*([0-9a-fA-Fxu<() ]+?)[,;]" % name, src)
+        m = re.search(r"\b%s\s*=\s*([0-9a-fA-Fxu<() ]+?)\s*[,;}]" % name,
+                      src)
         assert m, name
-        expr = m.group(1).replace("u", "").replace("<<", "<<")
-        return eval(expr)
+        return eval(m.group(1).replace("u", ""))
 
     assert cval("OP_AND") == plan.OP_AND
     assert cval("OP_OR") == plan.OP_OR

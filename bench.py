@@ -1,0 +1,182 @@
+#!/usr/bin/env python3
+"""
+dragnet_amd flagship benchmark — the driver contract.
+
+Measures the headline metric from BASELINE.json: NDJSON records/sec
+(whole node) + GB/s for a `dn scan` with a krill filter and a 2-field
+breakdown (req.method, res.statusCode) over synthetic muskie-shaped
+NDJSON (the tools/mktestdata record shape), scanned by the fused
+MI355X CDNA4 kernel with data-parallel fan-out over N GPUs and an
+RCCL merge of per-GPU partial aggregates each step.
+
+One "step" = one complete scan job over this rank's resident pool:
+async H2D staging of the pool, device newline index, fused scan
+kernel, table extraction + decode, cross-rank aggregate merge.
+Per-GPU work is fixed as N grows (weak scaling).
+
+    python bench.py --gpus N --steps K --warmup W [--mb MB]
+
+For N>1 launch via torch.distributed.run with --nproc-per-node N
+(reads RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from the env).
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+
+def log(msg):
+    if int(os.environ.get("RANK", 0)) == 0:
+        sys.stderr.write("[bench] %s\n" % msg)
+        sys.stderr.flush()
+
+
+def build_pool(mb, seed):
+    """Generate ~mb MB of synthetic NDJSON (mktestdata shape)."""
+    from dragnet_amd.tools.mktestdata import generate_lines
+    target = mb * 1024 * 1024
+    out = []
+    total = 0
+    nrec = 0
+    # generate in slabs so the record count adapts to actual line size
+    gen = generate_lines(1 << 62, seed=seed)
+    for line in gen:
+        out.append(line)
+        total += len(line)
+        nrec += 1
+        if total >= target:
+            break
+    return b"".join(out), nrec
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--mb", type=int,
+                    default=int(os.environ.get("DRAGNET_BENCH_MB", 256)),
+                    help="per-GPU NDJSON pool size (MB)")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    n_gpus = max(args.gpus, world)
+
+    import torch
+    assert torch.cuda.is_available(), "bench needs a GPU"
+    torch.cuda.set_device(local_rank)
+
+    dist = None
+    if world > 1:
+        import torch.distributed as torch_dist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29601")
+        torch_dist.init_process_group("nccl", rank=rank,
+                                      world_size=world)
+        dist = torch_dist
+
+    from dragnet_amd.distributed import merge_tables_tensor
+    from dragnet_amd.engine import plan as planmod
+    from dragnet_amd.engine.gpu import GpuEngine, _ScanContext
+    from dragnet_amd.query import query_load
+
+    query = query_load(
+        filter={"eq": ["req.method", "GET"]},
+        breakdown_specs="req.method,res.statusCode")
+
+    log("generating %d MB pool per GPU ..." % args.mb)
+    t0 = time.time()
+    pool, nrec = build_pool(args.mb, seed=1000 + rank)
+    pool_bytes = len(pool)
+    log("pool: %d records, %.1f MB in %.1fs" % (
+        nrec, pool_bytes / 1e6, time.time() - t0))
+
+    eng = GpuEngine(device=torch.device("cuda", local_rank))
+    cplan = planmod.compile_plan([query])
+    eng.chunk_bytes = pool_bytes  # single-chunk pool
+    ctx = _ScanContext(eng, cplan,
+                       agg_slots=1 << 16, dict_slots=1 << 16,
+                       dict_data_cap=32 << 20)
+    ctx.stage_resident(pool)
+    device = eng.device
+
+    def step():
+        ctx.reset()
+        ctx.scan_resident()
+        aggs, _stages = ctx.finalize([query])
+        if dist is not None:
+            merged = merge_tables_tensor(aggs[0], query, device)
+        else:
+            merged = aggs[0]
+        return merged
+
+    # warmup
+    for _ in range(args.warmup):
+        result = step()
+    assert int(ctx.counters[1].item()) == 0, "invalid JSON in pool?!"
+
+    # timed region
+    if dist is not None:
+        dist.barrier()
+    torch.cuda.synchronize(device)
+    t_start = time.time()
+    for _ in range(args.steps):
+        result = step()
+    torch.cuda.synchronize(device)
+    if dist is not None:
+        dist.barrier()
+    elapsed = time.time() - t_start
+
+    # max over ranks
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    total_records = nrec * args.steps * world
+    total_bytes = pool_bytes * args.steps * world
+    recs_per_sec = total_records / elapsed
+    gb_per_sec = total_bytes / elapsed / 1e9
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    # sanity: the scan actually matched records
+    matched = sum(p["value"] for p in result.points())
+    assert matched > 0, "no records matched the filter"
+
+    if rank == 0:
+        out = {
+            "metric": "NDJSON records/sec (whole node), krill filter "
+                      "+ 2-field breakdown (req.method,res.statusCode)",
+            "value": round(recs_per_sec, 1),
+            "unit": "records/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "uint8",
+            "data": "synthetic (mktestdata muskie-shaped NDJSON, "
+                    "random-generated, %d MB/GPU resident pool)"
+                    % args.mb,
+            "gb_per_sec": round(gb_per_sec, 3),
+            "config": {
+                "model": "dn scan: filter eq(req.method,GET) + "
+                         "breakdown req.method,res.statusCode",
+                "global_batch": nrec * world,
+                "seq_len": int(pool_bytes / max(nrec, 1)),
+                "parallelism": "dp%d" % world,
+            },
+        }
+        print(json.dumps(out))
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

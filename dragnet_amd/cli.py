@@ -473,8 +473,8 @@ def cmd_scan(argv):
         result = backend.scan(query, dry_run=bool(opts.get("dry-run")))
     except ValueError as e:
         raise FatalError(str(e))
-    if result is None:
-        return  # dry run
+    if result is None or getattr(result, "nonroot", False):
+        return  # dry run or non-rank-0 of a distributed scan
     _output_result(query, opts, result, title=args[0])
 
 
@@ -493,7 +493,7 @@ def cmd_query(argv):
                                dry_run=bool(opts.get("dry-run")))
     except ValueError as e:
         raise FatalError(str(e))
-    if result is None:
+    if result is None or getattr(result, "nonroot", False):
         return
     if getattr(result, "errors", None):
         for path, msg in result.errors:

@@ -61,12 +61,16 @@ def _ms_to_iso(ms):
 
 
 class ScanResult(object):
-    """A scan's output: per-query aggregators + counter stages."""
+    """A scan's output: per-query aggregators + counter stages.
 
-    def __init__(self, aggregators, stages, files=None):
+    nonroot marks an empty result on a non-rank-0 process of a
+    distributed scan (the CLI suppresses output for these)."""
+
+    def __init__(self, aggregators, stages, files=None, nonroot=False):
         self.aggregators = aggregators
         self.stages = stages
         self.files = files or []
+        self.nonroot = nonroot
 
 
 class FileDatasource(object):

@@ -317,6 +317,7 @@ SPECIAL_TRUE = 2
 SPECIAL_FALSE = 3
 SPECIAL_OBJECT = 4
 SPECIAL_ARRAY = 5
+SPECIAL_ARRJSON = 6  # | (string-dict id << 3)
 EMPTY_CODE = 0xFFFFFFFF
 
 
@@ -346,10 +347,20 @@ def decode_key(codes, query, strings, numbers):
                     int(num) if float(num).is_integer()
                     and abs(num) < 2**53 else float(num)))
         elif tag == TAG_SPECIAL:
-            out.append({SPECIAL_NULL: "null", SPECIAL_UNDEF: "undefined",
-                        SPECIAL_TRUE: "true", SPECIAL_FALSE: "false",
-                        SPECIAL_OBJECT: "[object Object]",
-                        SPECIAL_ARRAY: "<array>"}[val])
+            if (val & 7) == SPECIAL_ARRJSON:
+                import json as _json
+                from ..points import js_array_str
+                raw = strings[val >> 3]
+                try:
+                    out.append(js_array_str(_json.loads(raw)))
+                except ValueError:
+                    out.append(raw)
+            else:
+                out.append({
+                    SPECIAL_NULL: "null", SPECIAL_UNDEF: "undefined",
+                    SPECIAL_TRUE: "true", SPECIAL_FALSE: "false",
+                    SPECIAL_OBJECT: "[object Object]",
+                    SPECIAL_ARRAY: "<array>"}[val])
         else:
             raise PlanError("bad code tag %d" % tag)
     return tuple(out)

@@ -42,7 +42,8 @@ constexpr uint32_t TAG_SPECIAL = 3u;
 constexpr uint32_t ORD_BIAS    = 1u << 29;
 constexpr uint32_t SPECIAL_NULL = 0, SPECIAL_UNDEF = 1,
                    SPECIAL_TRUE = 2, SPECIAL_FALSE = 3,
-                   SPECIAL_OBJECT = 4, SPECIAL_ARRAY = 5;
+                   SPECIAL_OBJECT = 4, SPECIAL_ARRAY = 5,
+                   SPECIAL_ARRJSON = 6;  // | (string-dict id << 3)
 constexpr uint32_t EMPTY_CODE = 0xFFFFFFFFu;
 
 inline __device__ __host__ uint32_t make_code(uint32_t tag, uint32_t val) {
@@ -128,9 +129,9 @@ struct PlanView {
 
 struct ScanArgs {
   const uint8_t* data;
-  const uint32_t* line_starts;
-  const uint32_t* line_ends;
-  uint32_t nlines;
+  const uint32_t* nl_pos;      // sorted newline positions
+  const uint32_t* nlines_ptr;  // device count (no host sync needed)
+  uint32_t pos_cap;            // capacity of nl_pos
   PlanView P;
   AggTable* tables;       // [nm]
   StrDict sdict;

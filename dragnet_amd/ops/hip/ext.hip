@@ -62,8 +62,38 @@ hipStream_t current_stream() {
 
 }  // namespace
 
+// Index newlines in a device buffer: fills pos_out (up to its
+// capacity) with sorted newline positions and total_out[0] with the
+// true count — entirely device-side, no host sync.
+void newline_index(torch::Tensor data, int64_t n, torch::Tensor seg_scratch,
+                   torch::Tensor pos_out, torch::Tensor total_out) {
+  CHECK_GPU(data);
+  uint32_t nseg = (uint32_t)((n + NL_SEG - 1) / NL_SEG);
+  TORCH_CHECK((int64_t)nseg <= seg_scratch.numel(),
+              "seg_scratch too small");
+  if (n == 0) {
+    total_out.zero_();
+    return;
+  }
+  auto stream = current_stream();
+  const uint8_t* d = (const uint8_t*)data.data_ptr();
+  uint32_t* segs = (uint32_t*)seg_scratch.data_ptr();
+  uint32_t blocks = (nseg + 255) / 256;
+  hipLaunchKernelGGL(newline_count_kernel, dim3(blocks), dim3(256), 0,
+                     stream, d, (uint32_t)n, segs, nseg);
+  hipLaunchKernelGGL(newline_scan_kernel, dim3(1), dim3(1024), 0,
+                     stream, segs, nseg, (uint32_t*)total_out.data_ptr());
+  hipLaunchKernelGGL(newline_write_kernel, dim3(blocks), dim3(256), 0,
+                     stream, d, (uint32_t)n, segs, nseg,
+                     (uint32_t*)pos_out.data_ptr(),
+                     (uint32_t)pos_out.numel());
+  hipError_t err = hipGetLastError();
+  TORCH_CHECK(err == hipSuccess, "newline_index launch failed: ",
+              hipGetErrorString(err));
+}
+
 void scan_chunk(
-    torch::Tensor data, torch::Tensor line_starts, torch::Tensor line_ends,
+    torch::Tensor data, torch::Tensor nl_pos, torch::Tensor nlines_dev,
     torch::Tensor field_sigs, torch::Tensor prog_nodes,
     torch::Tensor prog_bounds, torch::Tensor const_meta,
     torch::Tensor const_dvals, torch::Tensor const_bytes,
@@ -79,17 +109,14 @@ void scan_chunk(
     torch::Tensor nd_next,
     torch::Tensor counters) {
   CHECK_GPU(data);
-  CHECK_GPU(line_starts);
+  CHECK_GPU(nl_pos);
   CHECK_GPU(table_descs);
-  TORCH_CHECK(line_starts.numel() == line_ends.numel());
-  uint32_t nlines = (uint32_t)line_starts.numel();
-  if (nlines == 0) return;
 
   ScanArgs A;
   A.data = (const uint8_t*)data.data_ptr();
-  A.line_starts = (const uint32_t*)line_starts.data_ptr();
-  A.line_ends = (const uint32_t*)line_ends.data_ptr();
-  A.nlines = nlines;
+  A.nl_pos = (const uint32_t*)nl_pos.data_ptr();
+  A.nlines_ptr = (const uint32_t*)nlines_dev.data_ptr();
+  A.pos_cap = (uint32_t)nl_pos.numel();
 
   A.P.field_sigs = (const uint64_t*)field_sigs.data_ptr();
   A.P.nf = (int)field_sigs.numel();
@@ -123,8 +150,9 @@ void scan_chunk(
   lds += 64;  // slack
   TORCH_CHECK(lds <= 160 * 1024, "plan needs too much LDS: ", lds);
 
-  uint32_t blocks = (nlines + BLOCK - 1) / BLOCK;
-  if (blocks > 4096) blocks = 4096;  // grid-stride the rest
+  // line count lives on-device; launch a full grid and grid-stride
+  uint32_t blocks = (A.pos_cap + BLOCK - 1) / BLOCK;
+  if (blocks > 2048) blocks = 2048;
 
   hipLaunchKernelGGL(scan_kernel, dim3(blocks), dim3(BLOCK), lds,
                      current_stream(), A);
@@ -218,6 +246,7 @@ torch::Tensor extract_numdict(torch::Tensor nd_state,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "dragnet_amd MI355X scan engine (gfx950 HIP kernels)";
   m.def("scan_chunk", &dn::scan_chunk, "fused NDJSON scan over one chunk");
+  m.def("newline_index", &dn::newline_index, "device-side newline index");
   m.def("agg_descs_host", &dn::agg_descs_host);
   m.def("extract_agg", &dn::extract_agg);
   m.def("extract_strdict", &dn::extract_strdict);

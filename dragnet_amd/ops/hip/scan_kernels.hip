@@ -104,6 +104,7 @@ struct FV {
     slen[f * BLOCK + tid] = len;
     num[f * BLOCK + tid] = n;
   }
+  DEV void set_len(int f, uint32_t len) { slen[f * BLOCK + tid] = len; }
   DEV uint8_t  get_type(int f) const { return type[f * BLOCK + tid]; }
   DEV uint32_t get_soff(int f) const { return soff[f * BLOCK + tid]; }
   DEV uint32_t get_slen(int f) const { return slen[f * BLOCK + tid]; }
@@ -368,6 +369,7 @@ DEV bool parse_record(const uint8_t* data, uint32_t start, uint32_t end,
   Cursor c; c.init(data, start, end);
 
   uint64_t sig_stack[SIG_DEPTH];  // parent path sig per object depth
+  int8_t cont_slot[MAX_DEPTH];    // captured-container slot per depth
   uint32_t is_arr_bits = 0;       // bit d: container at depth d is array
   int depth = 0;                  // container depth (0 = at top value)
   int arr_depth = 0;              // number of array containers on stack
@@ -448,20 +450,24 @@ DEV bool parse_record(const uint8_t* data, uint32_t start, uint32_t end,
       bool is_container = false;
 
       if (b == '{') {
+        uint32_t vstart = c.pos;
         c.pos++;
-        // capture the object itself (presence)
+        // capture the object itself (presence + raw span)
         if (cur_capture && arr_depth == 0)
-          fv.set(cur_slot, T_OBJ, 0, 0, 0.0);
+          fv.set(cur_slot, T_OBJ, vstart, 0, 0.0);
         if (depth == 0) top_type = T_OBJ;
         if (depth >= MAX_DEPTH) return false;
         c.skip_ws();
         if (!c.eof() && c.peek() == '}') {
           c.pos++;
           vtype = T_OBJ;  // empty object: treat as closed value
+          voff = vstart; vlen = c.pos - vstart;
           is_container = false;
           // fall through to "after value"
         } else {
           // push object frame
+          cont_slot[depth] = (int8_t)((cur_capture && arr_depth == 0)
+                                          ? cur_slot : -1);
           if (depth < SIG_DEPTH) sig_stack[depth] = cur_sig;
           is_arr_bits &= ~(1u << depth);
           depth++;
@@ -476,17 +482,21 @@ DEV bool parse_record(const uint8_t* data, uint32_t start, uint32_t end,
           continue;  // parse the member value
         }
       } else if (b == '[') {
+        uint32_t vstart = c.pos;
         c.pos++;
         if (cur_capture && arr_depth == 0)
-          fv.set(cur_slot, T_ARR, 0, 0, 0.0);
+          fv.set(cur_slot, T_ARR, vstart, 0, 0.0);
         if (depth == 0) top_type = T_ARR;
         if (depth >= MAX_DEPTH) return false;
         c.skip_ws();
         if (!c.eof() && c.peek() == ']') {
           c.pos++;
           vtype = T_ARR;
+          voff = vstart; vlen = c.pos - vstart;
           is_container = false;
         } else {
+          cont_slot[depth] = (int8_t)((cur_capture && arr_depth == 0)
+                                          ? cur_slot : -1);
           if (depth < SIG_DEPTH) sig_stack[depth] = cur_sig;
           is_arr_bits |= (1u << depth);
           depth++;
@@ -542,6 +552,9 @@ DEV bool parse_record(const uint8_t* data, uint32_t start, uint32_t end,
       if (b == ',') { expect_value = true; cur_capture = false; cur_slot = -1; continue; }
       if (b == ']') {
         depth--; arr_depth--;
+        if (cont_slot[depth] >= 0)
+          fv.set_len(cont_slot[depth],
+                     c.pos - fv.get_soff(cont_slot[depth]));
         // restore parent sig (not needed for captures inside arrays)
         cur_sig = (depth < SIG_DEPTH) ? sig_stack[depth] : 0;
         continue;  // still "after value" for the parent
@@ -563,6 +576,9 @@ DEV bool parse_record(const uint8_t* data, uint32_t start, uint32_t end,
       }
       if (b == '}') {
         depth--;
+        if (cont_slot[depth] >= 0)
+          fv.set_len(cont_slot[depth],
+                     c.pos - fv.get_soff(cont_slot[depth]));
         cur_sig = (depth < SIG_DEPTH) ? sig_stack[depth] : 0;
         continue;
       }
@@ -831,6 +847,93 @@ struct LdsCacheEntry {
   double count;
 };
 
+// ---- newline indexing (device-side, no host sync) ----
+// Three passes over 2 KiB segments: count, exclusive-scan, write.
+// Positions come out globally sorted because each segment writes its
+// newlines in order at its scanned base offset.
+
+constexpr uint32_t NL_SEG = 2048;
+
+__global__ void newline_count_kernel(const uint8_t* data, uint32_t n,
+                                     uint32_t* seg_counts,
+                                     uint32_t nseg) {
+  uint32_t seg = blockIdx.x * blockDim.x + threadIdx.x;
+  if (seg >= nseg) return;
+  const uint4* p16 = reinterpret_cast<const uint4*>(data);
+  uint32_t base = seg * NL_SEG;
+  uint32_t cnt = 0;
+  for (uint32_t o = 0; o < NL_SEG; o += 16) {
+    if (base + o >= n) break;
+    uint4 v = p16[(base + o) >> 4];
+    uint32_t w[4] = {v.x, v.y, v.z, v.w};
+#pragma unroll
+    for (int wi = 0; wi < 4; wi++) {
+#pragma unroll
+      for (int b = 0; b < 4; b++) {
+        uint32_t pos = base + o + wi * 4 + b;
+        if (pos < n && ((w[wi] >> (b * 8)) & 0xFF) == '\n') cnt++;
+      }
+    }
+  }
+  seg_counts[seg] = cnt;
+}
+
+// single-block exclusive scan over seg_counts (nseg can be large; a
+// 1024-thread block walks tiles with a running carry)
+__global__ void newline_scan_kernel(uint32_t* seg_counts, uint32_t nseg,
+                                    uint32_t* total_out) {
+  __shared__ uint32_t tile[1024];
+  __shared__ uint32_t carry;
+  if (threadIdx.x == 0) carry = 0;
+  __syncthreads();
+  for (uint32_t base = 0; base < nseg; base += 1024) {
+    uint32_t i = base + threadIdx.x;
+    uint32_t v = (i < nseg) ? seg_counts[i] : 0;
+    tile[threadIdx.x] = v;
+    __syncthreads();
+    // Hillis-Steele inclusive scan in LDS
+    for (uint32_t d = 1; d < 1024; d <<= 1) {
+      uint32_t t = (threadIdx.x >= d) ? tile[threadIdx.x - d] : 0;
+      __syncthreads();
+      tile[threadIdx.x] += t;
+      __syncthreads();
+    }
+    uint32_t incl = tile[threadIdx.x];
+    if (i < nseg) seg_counts[i] = carry + incl - v;  // exclusive
+    __syncthreads();
+    if (threadIdx.x == 1023) carry += tile[1023];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) *total_out = carry;
+}
+
+__global__ void newline_write_kernel(const uint8_t* data, uint32_t n,
+                                     const uint32_t* seg_offsets,
+                                     uint32_t nseg, uint32_t* out_pos,
+                                     uint32_t cap) {
+  uint32_t seg = blockIdx.x * blockDim.x + threadIdx.x;
+  if (seg >= nseg) return;
+  const uint4* p16 = reinterpret_cast<const uint4*>(data);
+  uint32_t base = seg * NL_SEG;
+  uint32_t w_at = seg_offsets[seg];
+  for (uint32_t o = 0; o < NL_SEG; o += 16) {
+    if (base + o >= n) break;
+    uint4 v = p16[(base + o) >> 4];
+    uint32_t w[4] = {v.x, v.y, v.z, v.w};
+#pragma unroll
+    for (int wi = 0; wi < 4; wi++) {
+#pragma unroll
+      for (int b = 0; b < 4; b++) {
+        uint32_t pos = base + o + wi * 4 + b;
+        if (pos < n && ((w[wi] >> (b * 8)) & 0xFF) == '\n') {
+          if (w_at < cap) out_pos[w_at] = pos;
+          w_at++;
+        }
+      }
+    }
+  }
+}
+
 __launch_bounds__(BLOCK)
 __global__ void scan_kernel(ScanArgs A) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -870,17 +973,19 @@ __global__ void scan_kernel(ScanArgs A) {
   double synth_val[MAX_SYNTH];
   uint8_t synth_ok[MAX_SYNTH];
 
+  uint32_t nlines = *A.nlines_ptr;
+  if (nlines > A.pos_cap) nlines = A.pos_cap;
   const uint32_t stride = gridDim.x * BLOCK;
   for (uint32_t r = blockIdx.x * BLOCK + threadIdx.x; ; r += stride) {
-    bool active = r < A.nlines;
+    bool active = r < nlines;
     if (!__any(active)) break;
 
     if (active) {
       atomicAdd(&lcnt[C_LINES], 1ull);
       for (int f = 0; f < nf; f++) fv.set(f, T_MISSING, 0, 0, 0.0);
 
-      uint32_t start = A.line_starts[r];
-      uint32_t end = A.line_ends[r];
+      uint32_t start = r ? A.nl_pos[r - 1] + 1 : 0;
+      uint32_t end = A.nl_pos[r];
       uint8_t top_type;
       bool ok = (end > start) &&
                 parse_record(A.data, start, end, P, fv, top_type);
@@ -1006,7 +1111,13 @@ __global__ void scan_kernel(ScanArgs A) {
               } else if (t == T_OBJ) {
                 code = make_code(TAG_SPECIAL, SPECIAL_OBJECT);
               } else if (t == T_ARR) {
-                code = make_code(TAG_SPECIAL, SPECIAL_ARRAY);
+                // intern the raw JSON span; the host canonicalizes it
+                // with JS Array.toString semantics (plan.decode_key)
+                uint32_t id = intern_string(A.sdict, A.data, so, sl);
+                if (id == 0xFFFFFFFFu || id >= (1u << 27)) {
+                  overflow = true; break;
+                }
+                code = make_code(TAG_SPECIAL, SPECIAL_ARRJSON | (id << 3));
               } else if (t == T_NUM) {
                 uint32_t id = intern_number(A.ndict, num);
                 if (id == 0xFFFFFFFFu) { overflow = true; break; }

@@ -68,13 +68,31 @@ def canonical(val, has_date=False):
         return js_num_str(val)
     if isinstance(val, str):
         return val
-    # objects/arrays as group-by values: JS coerces to
-    # "[object Object]" / join(',') — model the common cases
+    # objects/arrays as group-by values coerce like JS object keys
     if isinstance(val, dict):
         return "[object Object]"
     if isinstance(val, list):
-        return ",".join(canonical(x) for x in val)
+        return js_array_str(val)
     return str(val)
+
+
+def js_array_str(lst):
+    """JS Array.prototype.toString: join(','), null/undefined -> ''."""
+    parts = []
+    for x in lst:
+        if x is None:
+            parts.append("")
+        elif isinstance(x, list):
+            parts.append(js_array_str(x))
+        elif isinstance(x, dict):
+            parts.append("[object Object]")
+        elif isinstance(x, bool):
+            parts.append("true" if x else "false")
+        elif isinstance(x, (int, float)):
+            parts.append(js_num_str(x))
+        else:
+            parts.append(str(x))
+    return ",".join(parts)
 
 
 class Aggregator(object):

@@ -218,10 +218,10 @@ class _ScanContext(object):
         if self._copy_ev is None:
             self._copy_ev = torch.cuda.Event()
         self._copy_ev.record()
-        self.eng.ops.newline_index(dev_data, n, self._segs, self._pos,
-                                   self._nlines)
+        self.eng.ops.newline_index(dev_data, 0, n, self._segs,
+                                   self._pos, self._nlines)
         self.eng.ops.scan_chunk(
-            dev_data, self._pos, self._nlines,
+            dev_data, self._pos, self._nlines, 0,
             self.field_sigs, self.prog_nodes, self.prog_bounds,
             self.const_meta, self.const_dvals, self.const_bytes,
             self.synth_slots, self.cplan.n_synth,
@@ -243,10 +243,12 @@ class _ScanContext(object):
 
     # ---- resident-pool path (bench / repeated scans) ----
 
-    def stage_resident(self, buf):
+    def stage_resident(self, buf, n_slices=6):
         """Stage a byte pool into the pinned buffer once; later
         scan_resident() calls re-run the H2D + kernels without the
-        host-side copy."""
+        host-side copy.  The pool is split into newline-aligned
+        slices so the H2D copy of slice k+1 overlaps the scan of
+        slice k (copy stream + events)."""
         torch = self.t
         n = len(buf)
         padded = (n + 15) & ~15
@@ -254,19 +256,52 @@ class _ScanContext(object):
         pin = self._pinned
         pin[:n] = torch.frombuffer(bytearray(buf), dtype=torch.uint8)
         pin[n:padded] = 10
+
+        # newline-aligned slice boundaries
+        bounds = [0]
+        for k in range(1, n_slices):
+            target = n * k // n_slices
+            cut = buf.rfind(b"\n", 0, target)
+            start = cut + 1 if cut >= 0 else 0
+            if start > bounds[-1]:
+                bounds.append(start)
+        bounds.append(n)
+        self._slices = [(bounds[i], bounds[i + 1])
+                        for i in range(len(bounds) - 1)
+                        if bounds[i + 1] > bounds[i]]
         self._resident = (n, padded)
+        self._copy_stream = torch.cuda.Stream(device=self.eng.device)
+        self._slice_evs = [torch.cuda.Event()
+                           for _ in self._slices]
 
     def scan_resident(self):
-        """One full streaming pass over the staged pool: async H2D,
-        newline index, fused scan."""
+        """One full streaming pass over the staged pool: sliced async
+        H2D on a copy stream overlapping newline-index + fused scan on
+        the compute stream."""
+        torch = self.t
         n, padded = self._resident
         dev_data = self._dev_data
-        dev_data[:padded].copy_(self._pinned[:padded],
-                                non_blocking=True)
-        self.eng.ops.newline_index(dev_data, n, self._segs, self._pos,
-                                   self._nlines)
+        pin = self._pinned
+        main = torch.cuda.current_stream(self.eng.device)
+        # enqueue all slice copies on the copy stream
+        self._copy_stream.wait_stream(main)  # tables reset ordering
+        with torch.cuda.stream(self._copy_stream):
+            for k, (s, e) in enumerate(self._slices):
+                s16 = s & ~15
+                e16 = min((e + 15) & ~15, padded)
+                dev_data[s16:e16].copy_(pin[s16:e16],
+                                        non_blocking=True)
+                self._slice_evs[k].record(self._copy_stream)
+        # kernels per slice on the compute stream
+        for k, (s, e) in enumerate(self._slices):
+            main.wait_event(self._slice_evs[k])
+            self.eng.ops.newline_index(dev_data, s, e, self._segs,
+                                       self._pos, self._nlines)
+            self._scan_call(dev_data, s)
+
+    def _scan_call(self, dev_data, first_start):
         self.eng.ops.scan_chunk(
-            dev_data, self._pos, self._nlines,
+            dev_data, self._pos, self._nlines, first_start,
             self.field_sigs, self.prog_nodes, self.prog_bounds,
             self.const_meta, self.const_dvals, self.const_bytes,
             self.synth_slots, self.cplan.n_synth,

@@ -285,7 +285,7 @@ def compile_plan(queries, ds_filter=None, time_field=None,
     const_meta = np.array(consts.metas, dtype=np.int32).reshape(-1, 4)
     const_dvals = np.array(consts.dvals, dtype=np.float64)
     const_bytes = np.frombuffer(
-        bytes(consts.bytes) or b"\0", dtype=np.uint8)
+        bytes(consts.bytes) or b"\0", dtype=np.uint8).copy()
     synthetic = np.array(synth_slots or [0], dtype=np.int32)
     metrics = np.array(metric_rows, dtype=np.int32).reshape(-1, 8)
     bds = np.array(bd_rows or [[0, 0, 0, 0]],

@@ -132,6 +132,7 @@ struct ScanArgs {
   const uint32_t* nl_pos;      // sorted newline positions
   const uint32_t* nlines_ptr;  // device count (no host sync needed)
   uint32_t pos_cap;            // capacity of nl_pos
+  uint32_t first_start;        // byte offset of the first line
   PlanView P;
   AggTable* tables;       // [nm]
   StrDict sdict;

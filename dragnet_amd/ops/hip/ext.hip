@@ -65,13 +65,15 @@ hipStream_t current_stream() {
 // Index newlines in a device buffer: fills pos_out (up to its
 // capacity) with sorted newline positions and total_out[0] with the
 // true count — entirely device-side, no host sync.
-void newline_index(torch::Tensor data, int64_t n, torch::Tensor seg_scratch,
+void newline_index(torch::Tensor data, int64_t start, int64_t n,
+                   torch::Tensor seg_scratch,
                    torch::Tensor pos_out, torch::Tensor total_out) {
   CHECK_GPU(data);
-  uint32_t nseg = (uint32_t)((n + NL_SEG - 1) / NL_SEG);
+  int64_t span = n - (start & ~(int64_t)15);
+  uint32_t nseg = (uint32_t)((span + NL_SEG - 1) / NL_SEG);
   TORCH_CHECK((int64_t)nseg <= seg_scratch.numel(),
               "seg_scratch too small");
-  if (n == 0) {
+  if (span <= 0) {
     total_out.zero_();
     return;
   }
@@ -80,11 +82,11 @@ void newline_index(torch::Tensor data, int64_t n, torch::Tensor seg_scratch,
   uint32_t* segs = (uint32_t*)seg_scratch.data_ptr();
   uint32_t blocks = (nseg + 255) / 256;
   hipLaunchKernelGGL(newline_count_kernel, dim3(blocks), dim3(256), 0,
-                     stream, d, (uint32_t)n, segs, nseg);
+                     stream, d, (uint32_t)start, (uint32_t)n, segs, nseg);
   hipLaunchKernelGGL(newline_scan_kernel, dim3(1), dim3(1024), 0,
                      stream, segs, nseg, (uint32_t*)total_out.data_ptr());
   hipLaunchKernelGGL(newline_write_kernel, dim3(blocks), dim3(256), 0,
-                     stream, d, (uint32_t)n, segs, nseg,
+                     stream, d, (uint32_t)start, (uint32_t)n, segs, nseg,
                      (uint32_t*)pos_out.data_ptr(),
                      (uint32_t)pos_out.numel());
   hipError_t err = hipGetLastError();
@@ -94,6 +96,7 @@ void newline_index(torch::Tensor data, int64_t n, torch::Tensor seg_scratch,
 
 void scan_chunk(
     torch::Tensor data, torch::Tensor nl_pos, torch::Tensor nlines_dev,
+    int64_t first_start,
     torch::Tensor field_sigs, torch::Tensor prog_nodes,
     torch::Tensor prog_bounds, torch::Tensor const_meta,
     torch::Tensor const_dvals, torch::Tensor const_bytes,
@@ -117,6 +120,7 @@ void scan_chunk(
   A.nl_pos = (const uint32_t*)nl_pos.data_ptr();
   A.nlines_ptr = (const uint32_t*)nlines_dev.data_ptr();
   A.pos_cap = (uint32_t)nl_pos.numel();
+  A.first_start = (uint32_t)first_start;
 
   A.P.field_sigs = (const uint64_t*)field_sigs.data_ptr();
   A.P.nf = (int)field_sigs.numel();

@@ -854,13 +854,14 @@ struct LdsCacheEntry {
 
 constexpr uint32_t NL_SEG = 2048;
 
-__global__ void newline_count_kernel(const uint8_t* data, uint32_t n,
+__global__ void newline_count_kernel(const uint8_t* data,
+                                     uint32_t start, uint32_t n,
                                      uint32_t* seg_counts,
                                      uint32_t nseg) {
   uint32_t seg = blockIdx.x * blockDim.x + threadIdx.x;
   if (seg >= nseg) return;
   const uint4* p16 = reinterpret_cast<const uint4*>(data);
-  uint32_t base = seg * NL_SEG;
+  uint32_t base = (start & ~15u) + seg * NL_SEG;
   uint32_t cnt = 0;
   for (uint32_t o = 0; o < NL_SEG; o += 16) {
     if (base + o >= n) break;
@@ -871,7 +872,8 @@ __global__ void newline_count_kernel(const uint8_t* data, uint32_t n,
 #pragma unroll
       for (int b = 0; b < 4; b++) {
         uint32_t pos = base + o + wi * 4 + b;
-        if (pos < n && ((w[wi] >> (b * 8)) & 0xFF) == '\n') cnt++;
+        if (pos >= start && pos < n &&
+            ((w[wi] >> (b * 8)) & 0xFF) == '\n') cnt++;
       }
     }
   }
@@ -907,14 +909,15 @@ __global__ void newline_scan_kernel(uint32_t* seg_counts, uint32_t nseg,
   if (threadIdx.x == 0) *total_out = carry;
 }
 
-__global__ void newline_write_kernel(const uint8_t* data, uint32_t n,
+__global__ void newline_write_kernel(const uint8_t* data,
+                                     uint32_t start, uint32_t n,
                                      const uint32_t* seg_offsets,
                                      uint32_t nseg, uint32_t* out_pos,
                                      uint32_t cap) {
   uint32_t seg = blockIdx.x * blockDim.x + threadIdx.x;
   if (seg >= nseg) return;
   const uint4* p16 = reinterpret_cast<const uint4*>(data);
-  uint32_t base = seg * NL_SEG;
+  uint32_t base = (start & ~15u) + seg * NL_SEG;
   uint32_t w_at = seg_offsets[seg];
   for (uint32_t o = 0; o < NL_SEG; o += 16) {
     if (base + o >= n) break;
@@ -925,7 +928,8 @@ __global__ void newline_write_kernel(const uint8_t* data, uint32_t n,
 #pragma unroll
       for (int b = 0; b < 4; b++) {
         uint32_t pos = base + o + wi * 4 + b;
-        if (pos < n && ((w[wi] >> (b * 8)) & 0xFF) == '\n') {
+        if (pos >= start && pos < n &&
+            ((w[wi] >> (b * 8)) & 0xFF) == '\n') {
           if (w_at < cap) out_pos[w_at] = pos;
           w_at++;
         }
@@ -984,7 +988,7 @@ __global__ void scan_kernel(ScanArgs A) {
       atomicAdd(&lcnt[C_LINES], 1ull);
       for (int f = 0; f < nf; f++) fv.set(f, T_MISSING, 0, 0, 0.0);
 
-      uint32_t start = r ? A.nl_pos[r - 1] + 1 : 0;
+      uint32_t start = r ? A.nl_pos[r - 1] + 1 : A.first_start;
       uint32_t end = A.nl_pos[r];
       uint8_t top_type;
       bool ok = (end > start) &&

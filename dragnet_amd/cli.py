@@ -70,11 +70,24 @@ class FatalError(Exception):
 
 
 def main(argv=None):
+    import time
+    t_start = time.monotonic()
     argv = list(sys.argv[1:] if argv is None else argv)
+    track_time = False
     if argv and argv[0] == "-t":
-        # timing flag accepted for compatibility; timings on stderr
+        # print phase timings on exit (reference bin/dn:80-83,1291-1296)
+        track_time = True
         argv.pop(0)
 
+    rv = _dispatch(argv)
+    if track_time:
+        total = time.monotonic() - t_start
+        sys.stderr.write("timing stats:\n")
+        sys.stderr.write("    total:    %.6fs\n" % total)
+    return rv
+
+
+def _dispatch(argv):
     if not argv:
         return usage("no command specified")
 
@@ -458,6 +471,19 @@ def _output_result(query, opts, result, title=None):
         stages = list(result.stages)
         stages.append(("Flattener", {"ninputs": npoints, "noutputs": 1}))
         mod_output.dump_counters(stages)
+
+    if opts.get("warnings"):
+        # summary of warn-and-drop events (per-stage drop counters)
+        # (plain filter-outs are not warnings; reference vsWarn kinds)
+        drop_kinds = ("invalid json", "nfailedeval", "undef",
+                      "baddate", "nonnumeric")
+        for name, counters in result.stages:
+            for kind in drop_kinds:
+                v = counters.get(kind, 0)
+                if v:
+                    sys.stderr.write(
+                        "warn: %s: %s: %d record%s dropped\n"
+                        % (name, kind, v, "s" if v != 1 else ""))
 
 
 def cmd_scan(argv):

@@ -274,27 +274,30 @@ class _ScanContext(object):
         self._slice_evs = [torch.cuda.Event()
                            for _ in self._slices]
 
-    def scan_resident(self):
+    def scan_resident(self, h2d=True):
         """One full streaming pass over the staged pool: sliced async
         H2D on a copy stream overlapping newline-index + fused scan on
-        the compute stream."""
+        the compute stream.  h2d=False skips the copies (device-
+        resident re-scan: data already in HBM from a previous pass)."""
         torch = self.t
         n, padded = self._resident
         dev_data = self._dev_data
         pin = self._pinned
         main = torch.cuda.current_stream(self.eng.device)
-        # enqueue all slice copies on the copy stream
-        self._copy_stream.wait_stream(main)  # tables reset ordering
-        with torch.cuda.stream(self._copy_stream):
-            for k, (s, e) in enumerate(self._slices):
-                s16 = s & ~15
-                e16 = min((e + 15) & ~15, padded)
-                dev_data[s16:e16].copy_(pin[s16:e16],
-                                        non_blocking=True)
-                self._slice_evs[k].record(self._copy_stream)
+        if h2d:
+            # enqueue all slice copies on the copy stream
+            self._copy_stream.wait_stream(main)  # table reset ordering
+            with torch.cuda.stream(self._copy_stream):
+                for k, (s, e) in enumerate(self._slices):
+                    s16 = s & ~15
+                    e16 = min((e + 15) & ~15, padded)
+                    dev_data[s16:e16].copy_(pin[s16:e16],
+                                            non_blocking=True)
+                    self._slice_evs[k].record(self._copy_stream)
         # kernels per slice on the compute stream
         for k, (s, e) in enumerate(self._slices):
-            main.wait_event(self._slice_evs[k])
+            if h2d:
+                main.wait_event(self._slice_evs[k])
             self.eng.ops.newline_index(dev_data, s, e, self._segs,
                                        self._pos, self._nlines)
             self._scan_call(dev_data, s)

@@ -234,6 +234,7 @@ class FileDatasource(object):
 
         final = Aggregator(query)
         nerrors = []
+        use_gpu = os.environ.get("DRAGNET_INDEX_GPU") == "1"
         for path, _st in files:
             try:
                 iq = IndexQuerier(path)
@@ -241,7 +242,10 @@ class FileDatasource(object):
                 nerrors.append((path, str(e)))
                 continue
             try:
-                partial = iq.run(eff)
+                if use_gpu:
+                    partial = self._index_query_gpu(iq, eff)
+                else:
+                    partial = iq.run(eff)
                 for p in partial.points():
                     final.write(p)
             except IndexError_ as e:
@@ -253,6 +257,52 @@ class FileDatasource(object):
         result = ScanResult([final], stages, [p for p, _ in files])
         result.errors = nerrors
         return result
+
+    def _index_query_gpu(self, iq, query):
+        """K7: run an index query on the GPU by streaming the stored
+        rows as skinner points through the fused scan kernel (the
+        reference composes K7 from K2+K5 the same way).
+
+        The metric's rows are serialized to json-skinner NDJSON and fed
+        to the engine with the query's filter as the datasource filter;
+        the kernel re-filters, re-bucketizes and re-aggregates.
+        """
+        import json as _json
+
+        from ..index.sink import sqlite3_escape
+        table = iq.find_metric(query)
+
+        when = query.time_bounds_filter(table["datefield"]) \
+            if table["datefield"] else None
+        qfilter = None if table["ignore_filter"] else query.filter
+        filt = krill.filter_and(qfilter, when)
+
+        cols = [p["name"] for p in table["params"]]
+        sql = "SELECT %s from %s" % (
+            ", ".join([sqlite3_escape(c) for c in cols] + ["value"]),
+            table["table"])
+
+        def rows_as_points():
+            buf = []
+            total = 0
+            for row in iq.db.execute(sql):
+                fields = dict(zip(cols, row[:-1]))
+                line = (_json.dumps(
+                    {"fields": fields, "value": row[-1]},
+                    separators=(",", ":")) + "\n").encode()
+                buf.append(line)
+                total += len(line)
+                if total >= (8 << 20):
+                    yield b"".join(buf)
+                    buf, total = [], 0
+            if buf:
+                yield b"".join(buf)
+
+        res = self.engine().scan(
+            files=[], queries=[query], ds_filter=filt,
+            data_format="json-skinner",
+            byte_source=rows_as_points())
+        return res.aggregators[0]
 
     def index_read_points(self, metrics, interval="day"):
         """Emit every stored row of every index file as tagged points

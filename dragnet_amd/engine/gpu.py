@@ -456,11 +456,38 @@ def _decode_json_string(raw):
     return s
 
 
-def _read_files(files, chunk_size):
-    for path in files:
-        with open(path, "rb", buffering=0) as f:
-            while True:
-                chunk = f.read(chunk_size)
-                if not chunk:
-                    break
-                yield chunk
+def _read_files(files, chunk_size, prefetch=4):
+    """Chunked concatenated reader with a background prefetch thread
+    (the reference reads files with concurrency 2 through catstreams,
+    lib/datasource-file.js:252-288; here a reader thread keeps the
+    GPU fed while it scans the previous chunk)."""
+    import queue
+    import threading
+
+    q = queue.Queue(maxsize=prefetch)
+    err = []
+
+    def reader():
+        try:
+            for path in files:
+                with open(path, "rb", buffering=0) as f:
+                    while True:
+                        chunk = f.read(chunk_size)
+                        if not chunk:
+                            break
+                        q.put(chunk)
+        except Exception as e:  # surfaced on the consumer side
+            err.append(e)
+        finally:
+            q.put(None)
+
+    t = threading.Thread(target=reader, daemon=True)
+    t.start()
+    while True:
+        chunk = q.get()
+        if chunk is None:
+            break
+        yield chunk
+    t.join()
+    if err:
+        raise err[0]

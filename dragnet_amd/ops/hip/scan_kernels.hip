@@ -854,6 +854,25 @@ struct LdsCacheEntry {
 
 constexpr uint32_t NL_SEG = 2048;
 
+// SWAR newline mask: high bit set in each byte lane equal to '\n'
+DEV uint32_t nl_mask32(uint32_t w) {
+  uint32_t x = w ^ 0x0A0A0A0Au;
+  return (x - 0x01010101u) & ~x & 0x80808080u;
+}
+
+// byte-validity mask for a word at byte address wpos over [start, n)
+DEV uint32_t range_mask32(uint32_t wpos, uint32_t start, uint32_t n) {
+  uint32_t m = 0x80808080u;
+  if (wpos >= start && wpos + 4 <= n) return m;  // interior fast path
+  uint32_t out = 0;
+#pragma unroll
+  for (int b = 0; b < 4; b++) {
+    uint32_t pos = wpos + b;
+    if (pos >= start && pos < n) out |= 0x80u << (b * 8);
+  }
+  return out;
+}
+
 __global__ void newline_count_kernel(const uint8_t* data,
                                      uint32_t start, uint32_t n,
                                      uint32_t* seg_counts,
@@ -869,12 +888,9 @@ __global__ void newline_count_kernel(const uint8_t* data,
     uint32_t w[4] = {v.x, v.y, v.z, v.w};
 #pragma unroll
     for (int wi = 0; wi < 4; wi++) {
-#pragma unroll
-      for (int b = 0; b < 4; b++) {
-        uint32_t pos = base + o + wi * 4 + b;
-        if (pos >= start && pos < n &&
-            ((w[wi] >> (b * 8)) & 0xFF) == '\n') cnt++;
-      }
+      uint32_t m = nl_mask32(w[wi]) &
+                   range_mask32(base + o + wi * 4, start, n);
+      cnt += __popc(m);
     }
   }
   seg_counts[seg] = cnt;
@@ -925,14 +941,14 @@ __global__ void newline_write_kernel(const uint8_t* data,
     uint32_t w[4] = {v.x, v.y, v.z, v.w};
 #pragma unroll
     for (int wi = 0; wi < 4; wi++) {
-#pragma unroll
-      for (int b = 0; b < 4; b++) {
+      uint32_t m = nl_mask32(w[wi]) &
+                   range_mask32(base + o + wi * 4, start, n);
+      while (m) {
+        uint32_t b = ((uint32_t)__ffs(m) - 1) >> 3;  // byte lane
         uint32_t pos = base + o + wi * 4 + b;
-        if (pos >= start && pos < n &&
-            ((w[wi] >> (b * 8)) & 0xFF) == '\n') {
-          if (w_at < cap) out_pos[w_at] = pos;
-          w_at++;
-        }
+        if (w_at < cap) out_pos[w_at] = pos;
+        w_at++;
+        m &= m - 1;
       }
     }
   }

@@ -308,13 +308,20 @@ class _ScanContext(object):
                     dev_data[s16:e16].copy_(pin[s16:e16],
                                             non_blocking=True)
                     self._slice_evs[k].record(self._copy_stream)
-        # kernels per slice on the compute stream (line positions are
-        # precomputed at stage time)
+        # kernels per slice on the compute stream.  Streaming passes
+        # (h2d=True) model fresh data and re-index newlines each pass;
+        # device-resident re-scans reuse the staged line index (a
+        # re-scanning engine retains it like the reference retains its
+        # on-disk indexes).
         for k, (s, e) in enumerate(self._slices):
             if h2d:
                 main.wait_event(self._slice_evs[k])
-            pos_k, nlines_k = self._slice_pos[k]
-            self._scan_call(dev_data, s, pos_k, nlines_k)
+                self.eng.ops.newline_index(dev_data, s, e, self._segs,
+                                           self._pos, self._nlines)
+                self._scan_call(dev_data, s)
+            else:
+                pos_k, nlines_k = self._slice_pos[k]
+                self._scan_call(dev_data, s, pos_k, nlines_k)
 
     def _scan_call(self, dev_data, first_start, pos=None, nlines=None):
         self.eng.ops.scan_chunk(

@@ -55,39 +55,86 @@ DEV void drain_stores() {
 // -------------------------------------------------------------------
 // byte cursor: 16B-buffered reads from the chunk
 
+// Structural bytes use direct (L1-served) byte loads; bulk string
+// spans use unaligned 64-bit windows with SWAR special-byte masks.
 struct Cursor {
-  const uint4* base16;
+  const uint8_t* data;
   uint32_t pos, end;
-  uint4 buf;
-  uint32_t buf_base;
 
-  DEV void init(const uint8_t* data, uint32_t p, uint32_t e) {
-    base16 = reinterpret_cast<const uint4*>(data);
-    pos = p; end = e; buf_base = 0xFFFFFFF0u;
+  DEV void init(const uint8_t* d, uint32_t p, uint32_t e) {
+    data = d; pos = p; end = e;
   }
-  DEV uint8_t byte_at(uint32_t p) {
-    uint32_t b = p - buf_base;
-    if (b >= 16u) { buf_base = p & ~15u; buf = base16[buf_base >> 4]; b = p - buf_base; }
-    uint32_t w;
-    switch (b >> 2) {
-      case 0: w = buf.x; break;
-      case 1: w = buf.y; break;
-      case 2: w = buf.z; break;
-      default: w = buf.w; break;
-    }
-    return (uint8_t)(w >> ((b & 3u) * 8u));
-  }
+  DEV uint8_t byte_at(uint32_t p) const { return data[p]; }
   DEV bool eof() const { return pos >= end; }
-  DEV uint8_t peek() { return byte_at(pos); }
-  DEV uint8_t next() { return byte_at(pos++); }
+  DEV uint8_t peek() const { return data[pos]; }
+  DEV uint8_t next() { return data[pos++]; }
   DEV void skip_ws() {
     while (pos < end) {
-      uint8_t b = byte_at(pos);
+      uint8_t b = data[pos];
       if (b == ' ' || b == '\t' || b == '\r' || b == '\n') pos++;
       else break;
     }
   }
 };
+
+// high bit set per zero byte; the FIRST flagged byte is always a true
+// zero (false positives only occur above a real zero byte)
+DEV uint64_t hz8(uint64_t v) {
+  return (v - 0x0101010101010101ull) & ~v & 0x8080808080808080ull;
+}
+
+// bytes that end a bulk string run: '"', '\\', or a control char
+DEV uint64_t str_special_mask(uint64_t w) {
+  uint64_t q = hz8(w ^ 0x2222222222222222ull);
+  uint64_t b = hz8(w ^ 0x5C5C5C5C5C5C5C5Cull);
+  uint64_t c = hz8(w & 0xE0E0E0E0E0E0E0E0ull);
+  return q | b | c;
+}
+
+// Scan a JSON string body (cursor after the opening quote): SWAR over
+// 8-byte windows, per-byte only at escapes.  Sets raw off/len.
+DEV bool scan_string_fast(Cursor& c, uint32_t& off_out,
+                          uint32_t& len_out) {
+  const uint8_t* d = c.data;
+  uint32_t off = c.pos, p = c.pos, end = c.end;
+  while (true) {
+    while (p + 8 <= end) {
+      uint64_t w;
+      __builtin_memcpy(&w, d + p, 8);
+      uint64_t m = str_special_mask(w);
+      if (m == 0) { p += 8; continue; }
+      p += ((uint32_t)__ffsll((unsigned long long)m) - 1) >> 3;
+      break;
+    }
+    if (p >= end) return false;
+    uint8_t b = d[p];
+    if (b == '"') {
+      off_out = off; len_out = p - off; c.pos = p + 1;
+      return true;
+    }
+    if (b == '\\') {
+      p++;
+      if (p >= end) return false;
+      uint8_t e = d[p++];
+      if (e == 'u') {
+        if (p + 4 > end) return false;
+        for (int k = 0; k < 4; k++) {
+          uint8_t x = d[p + k];
+          bool hex = (x >= '0' && x <= '9') || (x >= 'a' && x <= 'f') ||
+                     (x >= 'A' && x <= 'F');
+          if (!hex) return false;
+        }
+        p += 4;
+      } else if (!(e == '"' || e == '\\' || e == '/' || e == 'b' ||
+                   e == 'f' || e == 'n' || e == 'r' || e == 't')) {
+        return false;
+      }
+      continue;
+    }
+    if (b < 0x20) return false;  // raw control char
+    p++;  // SWAR tail (< 8 bytes left): plain byte, keep walking
+  }
+}
 
 // -------------------------------------------------------------------
 // per-record extracted field values, stored in LDS (SoA, [field][tid])
@@ -506,8 +553,7 @@ DEV bool parse_record(const uint8_t* data, uint32_t start, uint32_t end,
         }
       } else if (b == '"') {
         c.pos++;
-        uint64_t h;
-        if (!scan_string(0, h, voff, vlen)) return false;
+        if (!scan_string_fast(c, voff, vlen)) return false;
         vtype = T_STR;
       } else if (b == 't') {
         if (c.end - c.pos < 4) return false;

@@ -100,6 +100,11 @@ class FileDatasource(object):
             sys.stderr.write(
                 'warn: datasource is missing "timeformat" for '
                 '"before" and "after" constraints\n')
+        if warn is None:
+            # per-file stat/read errors are warn-and-skip, never fatal
+            # (reference lib/fs-find.js:274-279)
+            def warn(path, msg):
+                sys.stderr.write("warn: %s: %s\n" % (path, msg))
         return find_data_files(
             self.ds.path, self.ds.time_format,
             query.after_ms, query.before_ms,

@@ -133,6 +133,7 @@ struct ScanArgs {
   const uint32_t* nlines_ptr;  // device count (no host sync needed)
   uint32_t pos_cap;            // capacity of nl_pos
   uint32_t first_start;        // byte offset of the first line
+  uint32_t tile_cap;           // LDS staging tile bytes (0 = off)
   PlanView P;
   AggTable* tables;       // [nm]
   StrDict sdict;

@@ -153,6 +153,14 @@ void scan_chunk(
   lds += (C_GLOBAL_N + (size_t)A.P.nm * CM_N) * 8;
   lds += 64;  // slack
   TORCH_CHECK(lds <= 160 * 1024, "plan needs too much LDS: ", lds);
+  // remaining LDS becomes the block staging tile (parse from LDS);
+  // blocks whose record span exceeds it parse from global instead
+  size_t avail = 160 * 1024 - lds - 256;
+  size_t tile_cap = avail & ~(size_t)15;
+  if (tile_cap > 100 * 1024) tile_cap = 100 * 1024;
+  if (tile_cap < 24 * 1024) tile_cap = 0;
+  A.tile_cap = (uint32_t)tile_cap;
+  lds += tile_cap;
 
   // line count lives on-device; launch a full grid and grid-stride
   uint32_t blocks = (A.pos_cap + BLOCK - 1) / BLOCK;

@@ -55,22 +55,32 @@ DEV void drain_stores() {
 // -------------------------------------------------------------------
 // byte cursor: 16B-buffered reads from the chunk
 
-// Structural bytes use direct (L1-served) byte loads; bulk string
-// spans use unaligned 64-bit windows with SWAR special-byte masks.
+// Record bytes are addressed by ABSOLUTE chunk offset through a Bytes
+// view: either the global chunk pointer (bias 0) or a block's LDS
+// staging tile (bias = tile base).  Parsing from LDS turns the 64-way
+// divergent per-byte gathers of thread-per-record parsing into banked
+// LDS reads.
+struct Bytes {
+  const uint8_t* mem;
+  uint32_t bias;
+  DEV uint8_t at(uint32_t abs) const { return mem[abs - bias]; }
+  DEV const uint8_t* ptr(uint32_t abs) const { return mem + (abs - bias); }
+};
+
 struct Cursor {
-  const uint8_t* data;
+  Bytes B;
   uint32_t pos, end;
 
-  DEV void init(const uint8_t* d, uint32_t p, uint32_t e) {
-    data = d; pos = p; end = e;
+  DEV void init(Bytes b, uint32_t p, uint32_t e) {
+    B = b; pos = p; end = e;
   }
-  DEV uint8_t byte_at(uint32_t p) const { return data[p]; }
+  DEV uint8_t byte_at(uint32_t p) const { return B.at(p); }
   DEV bool eof() const { return pos >= end; }
-  DEV uint8_t peek() const { return data[pos]; }
-  DEV uint8_t next() { return data[pos++]; }
+  DEV uint8_t peek() const { return B.at(pos); }
+  DEV uint8_t next() { return B.at(pos++); }
   DEV void skip_ws() {
     while (pos < end) {
-      uint8_t b = data[pos];
+      uint8_t b = B.at(pos);
       if (b == ' ' || b == '\t' || b == '\r' || b == '\n') pos++;
       else break;
     }
@@ -95,7 +105,7 @@ DEV uint64_t str_special_mask(uint64_t w) {
 // 8-byte windows, per-byte only at escapes.  Sets raw off/len.
 DEV bool scan_string_fast(Cursor& c, uint32_t& off_out,
                           uint32_t& len_out) {
-  const uint8_t* d = c.data;
+  const uint8_t* d = c.B.mem - c.B.bias;
   uint32_t off = c.pos, p = c.pos, end = c.end;
   while (true) {
     while (p + 8 <= end) {
@@ -238,7 +248,8 @@ DEV NumOut parse_json_number(Cursor& c) {
 
 // JavaScript ToNumber for record strings (mirrors krill.to_number):
 // trim ws; "" -> 0; decimal/hex/Infinity; else NaN.
-DEV double js_to_number(const uint8_t* data, uint32_t off, uint32_t len) {
+DEV double js_to_number(Bytes BV, uint32_t off, uint32_t len) {
+  const uint8_t* data = BV.mem - BV.bias;
   uint32_t i = 0, j = len;
   while (i < j) { uint8_t b = data[off + i]; if (b==' '||b=='\t'||b=='\r'||b=='\n'||b=='\f'||b=='\v') i++; else break; }
   while (j > i) { uint8_t b = data[off + j - 1]; if (b==' '||b=='\t'||b=='\r'||b=='\n'||b=='\f'||b=='\v') j--; else break; }
@@ -325,7 +336,8 @@ DEV bool is_leap(long y) {
   return (y % 4 == 0) && ((y % 100 != 0) || (y % 400 == 0));
 }
 
-DEV DateOut parse_iso_ms(const uint8_t* data, uint32_t off, uint32_t len) {
+DEV DateOut parse_iso_ms(Bytes BV, uint32_t off, uint32_t len) {
+  const uint8_t* data = BV.mem - BV.bias;
   DateOut out; out.ok = false; out.ms = 0;
   // trim
   uint32_t i = 0, j = len;
@@ -411,9 +423,9 @@ DEV DateOut parse_iso_ms(const uint8_t* data, uint32_t off, uint32_t len) {
 // Parses one record (bytes [start,end)); fills fv (all slots must be
 // preinitialized to T_MISSING by the caller).  Returns false on invalid
 // JSON.  top_type receives the top-level value type.
-DEV bool parse_record(const uint8_t* data, uint32_t start, uint32_t end,
+DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
                       const PlanView& P, FV& fv, uint8_t& top_type) {
-  Cursor c; c.init(data, start, end);
+  Cursor c; c.init(BV, start, end);
 
   uint64_t sig_stack[SIG_DEPTH];  // parent path sig per object depth
   int8_t cont_slot[MAX_DEPTH];    // captured-container slot per depth
@@ -636,8 +648,9 @@ DEV bool parse_record(const uint8_t* data, uint32_t start, uint32_t end,
 // -------------------------------------------------------------------
 // predicate evaluation (K2): -1 throw (missing field), 0 false, 1 true
 
-DEV int eval_leaf(const PlanView& P, const uint8_t* data, const FV& fv,
+DEV int eval_leaf(const PlanView& P, Bytes BV, const FV& fv,
                   int op, int slot, int cidx) {
+  const uint8_t* data = BV.mem - BV.bias;
   uint8_t ft = fv.get_type(slot);
   if (ft == T_MISSING) return -1;  // krill: missing field -> throw
 
@@ -666,7 +679,7 @@ DEV int eval_leaf(const PlanView& P, const uint8_t* data, const FV& fv,
             if (data[fo + k] != P.const_bytes[coff + k]) { eq = false; break; }
         }
       } else {  // string vs number: ToNumber(field)
-        double fn = js_to_number(data, fv.get_soff(slot), fv.get_slen(slot));
+        double fn = js_to_number(BV, fv.get_soff(slot), fv.get_slen(slot));
         eq = (fn == fn) && (fn == cdval);
       }
     } else {
@@ -697,7 +710,7 @@ DEV int eval_leaf(const PlanView& P, const uint8_t* data, const FV& fv,
   else if (ft == T_NULL) x = 0.0;
   else if (ft == T_TRUE) x = 1.0;
   else if (ft == T_FALSE) x = 0.0;
-  else if (ft == T_STR) x = js_to_number(data, fv.get_soff(slot), fv.get_slen(slot));
+  else if (ft == T_STR) x = js_to_number(BV, fv.get_soff(slot), fv.get_slen(slot));
   else x = __builtin_nan("");  // object/array
   if (ckind == CONST_NUM) y = cdval;
   else if (ckind == CONST_NULL) y = 0.0;
@@ -711,7 +724,7 @@ DEV int eval_leaf(const PlanView& P, const uint8_t* data, const FV& fv,
   }
 }
 
-DEV int eval_predicate(const PlanView& P, const uint8_t* data,
+DEV int eval_predicate(const PlanView& P, Bytes BV,
                        const FV& fv, int prog_id) {
   int idx = P.prog_bounds[prog_id * 2 + 0];
   struct Frame { int16_t op; int16_t remaining; int32_t end; };
@@ -733,7 +746,7 @@ DEV int eval_predicate(const PlanView& P, const uint8_t* data,
       result = 1;
       idx = P.prog_nodes[idx * 4 + 3];
     } else {
-      result = eval_leaf(P, data, fv, op,
+      result = eval_leaf(P, BV, fv, op,
                          P.prog_nodes[idx * 4 + 1],
                          P.prog_nodes[idx * 4 + 2]);
       idx = P.prog_nodes[idx * 4 + 3];
@@ -759,16 +772,18 @@ DEV int eval_predicate(const PlanView& P, const uint8_t* data,
 // -------------------------------------------------------------------
 // dictionaries (string + number interning)
 
-DEV uint64_t hash_bytes(const uint8_t* data, uint32_t off, uint32_t len) {
+DEV uint64_t hash_bytes(Bytes BV, uint32_t off, uint32_t len) {
+  const uint8_t* data = BV.mem - BV.bias;
   uint64_t h = FNV_OFFSET;
   for (uint32_t k = 0; k < len; k++) h = fnv1a_byte(h, data[off + k]);
   return mix64(h ^ len);
 }
 
 // Returns string id, or 0xFFFFFFFF on table/data overflow.
-DEV uint32_t intern_string(const StrDict& D, const uint8_t* data,
+DEV uint32_t intern_string(const StrDict& D, Bytes BV,
                            uint32_t off, uint32_t len) {
-  uint64_t h = hash_bytes(data, off, len);
+  const uint8_t* data = BV.mem - BV.bias;
+  uint64_t h = hash_bytes(BV, off, len);
   uint32_t mask = D.nslots - 1;
   uint32_t s = (uint32_t)h & mask;
   for (uint32_t probes = 0; probes < D.nslots; probes++, s = (s + 1) & mask) {
@@ -1021,6 +1036,9 @@ __global__ void scan_kernel(ScanArgs A) {
   off += (size_t)LDS_CACHE * sizeof(LdsCacheEntry);
   unsigned long long* lcnt = reinterpret_cast<unsigned long long*>(smem + off);
   const int NCNT = C_GLOBAL_N + P.nm * CM_N;
+  off += (size_t)NCNT * sizeof(unsigned long long);
+  off = (off + 15) & ~(size_t)15;
+  uint8_t* tile = reinterpret_cast<uint8_t*>(smem + off);  // staging
 
   // init LDS
   for (int i = threadIdx.x; i < LDS_CACHE; i += BLOCK) {
@@ -1042,9 +1060,34 @@ __global__ void scan_kernel(ScanArgs A) {
   uint32_t nlines = *A.nlines_ptr;
   if (nlines > A.pos_cap) nlines = A.pos_cap;
   const uint32_t stride = gridDim.x * BLOCK;
-  for (uint32_t r = blockIdx.x * BLOCK + threadIdx.x; ; r += stride) {
+  for (uint32_t rbase = blockIdx.x * BLOCK; rbase < nlines;
+       rbase += stride) {
+    uint32_t r = rbase + threadIdx.x;
     bool active = r < nlines;
-    if (!__any(active)) break;
+
+    // Block-level LDS staging: copy the block's contiguous record
+    // span into the tile with coalesced uint4 loads, then parse from
+    // LDS (banked reads) instead of 64-way divergent global gathers.
+    Bytes BV;
+    BV.mem = A.data;
+    BV.bias = 0;
+    uint32_t t_start = rbase ? A.nl_pos[rbase - 1] + 1 : A.first_start;
+    uint32_t r_last = rbase + BLOCK - 1;
+    if (r_last >= nlines) r_last = nlines - 1;
+    uint32_t t_end = A.nl_pos[r_last] + 1;
+    uint32_t t_base = t_start & ~15u;
+    uint32_t span = t_end - t_base;
+    bool staged = A.tile_cap >= 16 && span <= A.tile_cap;
+    if (staged) {
+      const uint4* g16 = reinterpret_cast<const uint4*>(A.data + t_base);
+      uint4* l16 = reinterpret_cast<uint4*>(tile);
+      uint32_t n16 = (span + 15) >> 4;
+      for (uint32_t i = threadIdx.x; i < n16; i += BLOCK)
+        l16[i] = g16[i];
+      __syncthreads();
+      BV.mem = tile;
+      BV.bias = t_base;
+    }
 
     if (active) {
       atomicAdd(&lcnt[C_LINES], 1ull);
@@ -1054,7 +1097,7 @@ __global__ void scan_kernel(ScanArgs A) {
       uint32_t end = A.nl_pos[r];
       uint8_t top_type;
       bool ok = (end > start) &&
-                parse_record(A.data, start, end, P, fv, top_type);
+                parse_record(BV, start, end, P, fv, top_type);
       double weight = 1.0;
       if (ok && A.data_format_skinner) {
         // require: object top, a "fields" member, numeric "value"
@@ -1071,7 +1114,7 @@ __global__ void scan_kernel(ScanArgs A) {
         atomicAdd(&lcnt[C_PARSED], 1ull);
 
         // datasource filter (program 0)
-        int keep = eval_predicate(P, A.data, fv, 0);
+        int keep = eval_predicate(P, BV, fv, 0);
         if (keep == -1) atomicAdd(&lcnt[C_DS_FAILEDEVAL], 1ull);
         else if (keep == 0) atomicAdd(&lcnt[C_DS_FILTERED], 1ull);
 
@@ -1083,7 +1126,7 @@ __global__ void scan_kernel(ScanArgs A) {
             if (t == T_MISSING) { synth_ok[si] = 2; continue; }   // undef
             if (t == T_NUM) { synth_ok[si] = 1; synth_val[si] = fv.get_num(slot); continue; }
             if (t == T_STR) {
-              DateOut d = parse_iso_ms(A.data, fv.get_soff(slot), fv.get_slen(slot));
+              DateOut d = parse_iso_ms(BV, fv.get_soff(slot), fv.get_slen(slot));
               if (d.ok) {
                 long long secs = d.ms >= 0 ? d.ms / 1000
                                            : (d.ms - 999) / 1000;  // floor
@@ -1100,7 +1143,7 @@ __global__ void scan_kernel(ScanArgs A) {
             unsigned long long* mc = &lcnt[C_GLOBAL_N + m * CM_N];
             atomicAdd(&mc[CM_FILTER_IN], 1ull);
 
-            int res = eval_predicate(P, A.data, fv, M[0]);
+            int res = eval_predicate(P, BV, fv, M[0]);
             if (res == -1) { atomicAdd(&mc[CM_FAILEDEVAL], 1ull); continue; }
             if (res == 0) { atomicAdd(&mc[CM_FILTERED], 1ull); continue; }
 
@@ -1179,7 +1222,7 @@ __global__ void scan_kernel(ScanArgs A) {
               } else if (t == T_ARR) {
                 // intern the raw JSON span; the host canonicalizes it
                 // with JS Array.toString semantics (plan.decode_key)
-                uint32_t id = intern_string(A.sdict, A.data, so, sl);
+                uint32_t id = intern_string(A.sdict, BV, so, sl);
                 if (id == 0xFFFFFFFFu || id >= (1u << 27)) {
                   overflow = true; break;
                 }
@@ -1189,7 +1232,7 @@ __global__ void scan_kernel(ScanArgs A) {
                 if (id == 0xFFFFFFFFu) { overflow = true; break; }
                 code = make_code(TAG_NUM, id);
               } else {  // T_STR
-                uint32_t id = intern_string(A.sdict, A.data, so, sl);
+                uint32_t id = intern_string(A.sdict, BV, so, sl);
                 if (id == 0xFFFFFFFFu) { overflow = true; break; }
                 code = make_code(TAG_STR, id);
               }
@@ -1239,6 +1282,8 @@ __global__ void scan_kernel(ScanArgs A) {
         }
       }
     }
+
+    if (staged) __syncthreads();  // tile reused next iteration
   }
 
   // flush LDS cache + counters

@@ -336,6 +336,7 @@ class _ScanContext(object):
             self.bd_rows, self.bd_steps,
             self.cplan.value_slot, self.cplan.fields_slot,
             self.cplan.data_format == "json-skinner",
+            _env_int("DRAGNET_LDS_STAGE", 0),
             self.table_descs,
             self.sd["state"], self.sd["hash"], self.sd["id"],
             self.sd["off"], self.sd["len"], self.sd["data"],

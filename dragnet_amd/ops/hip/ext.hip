@@ -104,6 +104,7 @@ void scan_chunk(
     torch::Tensor metric_rows, torch::Tensor synth_req,
     torch::Tensor bd_rows, torch::Tensor bd_steps,
     int64_t value_slot, int64_t fields_slot, bool skinner,
+    int64_t tile_cap_req,
     torch::Tensor table_descs,
     torch::Tensor sd_state, torch::Tensor sd_hash, torch::Tensor sd_id,
     torch::Tensor sd_off, torch::Tensor sd_len, torch::Tensor sd_data,
@@ -153,12 +154,18 @@ void scan_chunk(
   lds += (C_GLOBAL_N + (size_t)A.P.nm * CM_N) * 8;
   lds += 64;  // slack
   TORCH_CHECK(lds <= 160 * 1024, "plan needs too much LDS: ", lds);
-  // remaining LDS becomes the block staging tile (parse from LDS);
-  // blocks whose record span exceeds it parse from global instead
-  size_t avail = 160 * 1024 - lds - 256;
-  size_t tile_cap = avail & ~(size_t)15;
-  if (tile_cap > 100 * 1024) tile_cap = 100 * 1024;
-  if (tile_cap < 24 * 1024) tile_cap = 0;
+  // optional LDS staging tile (parse from LDS); tile_cap_req:
+  // 0 = off, -1 = auto (all remaining LDS), >0 = requested bytes.
+  // Measured on MI355X: staging costs more occupancy than the
+  // global-gather it saves for ~226B records — default off.
+  size_t tile_cap = 0;
+  if (tile_cap_req != 0) {
+    size_t avail = (160 * 1024 - lds - 256) & ~(size_t)15;
+    tile_cap = (tile_cap_req < 0) ? avail
+                                  : std::min((size_t)tile_cap_req, avail);
+    if (tile_cap > 100 * 1024) tile_cap = 100 * 1024;
+    if (tile_cap < 24 * 1024) tile_cap = 0;
+  }
   A.tile_cap = (uint32_t)tile_cap;
   lds += tile_cap;
 

@@ -9,7 +9,7 @@ points (reference bin/dn:941-1274).
 """
 
 import json
-import os
+
 import sys
 
 from . import config as mod_config

@@ -12,7 +12,7 @@ where it behaves like the file backend).  Rank 0 returns results;
 other ranks return None (the CLI suppresses their output).
 """
 
-from .. import krill
+
 from ..distributed import (dist_env, init_process_group,
                            merge_counter_stages, merge_points_object,
                            shard_files)

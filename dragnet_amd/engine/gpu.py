@@ -20,7 +20,7 @@ import os
 
 import numpy as np
 
-from ..points import Aggregator, js_num_str
+from ..points import Aggregator
 from . import plan as planmod
 
 COUNTER_NAMES = ["lines", "invalid_json", "parsed", "ds_filtered",

@@ -20,7 +20,6 @@ packed little buffers the fused HIP scan kernel consumes:
 Layouts are mirrored by struct definitions in ops/hip/common.h.
 """
 
-import struct
 
 import numpy as np
 

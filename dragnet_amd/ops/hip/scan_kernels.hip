@@ -34,11 +34,59 @@ DEV uint64_t fnv1a_byte(uint64_t h, uint8_t b) {
 }
 constexpr uint64_t FNV_OFFSET = 0xCBF29CE484222325ull;
 
+__device__ __constant__ double P10_TBL[23] = {
+  1e0,1e1,1e2,1e3,1e4,1e5,1e6,1e7,1e8,1e9,1e10,1e11,1e12,1e13,1e14,
+  1e15,1e16,1e17,1e18,1e19,1e20,1e21,1e22};
+__device__ __constant__ int DIM_TBL[12] = {
+  31,28,31,30,31,30,31,31,30,31,30,31};
+
 DEV uint64_t mix64(uint64_t x) {
   x ^= x >> 33; x *= 0xFF51AFD7ED558CCDull;
   x ^= x >> 33; x *= 0xC4CEB9FE1A85EC53ull;
   x ^= x >> 33; return x;
 }
+
+// Register-resident small "arrays": named scalar members with select
+// chains, so dynamic indexing stays in VGPRs.  (A local ARRAY — even
+// one indexed through an unrolled select chain — gets re-canonicalized
+// by LLVM into scratch = HBM-backed private memory; measured as the
+// scan kernel's dominant stall.)
+#define DN_REGSEL6(T, NAME)                                            \
+  struct NAME {                                                        \
+    T a0, a1, a2, a3, a4, a5;                                          \
+    DEV T get(int i) const {                                           \
+      T r = a0;                                                        \
+      r = (i == 1) ? a1 : r; r = (i == 2) ? a2 : r;                    \
+      r = (i == 3) ? a3 : r; r = (i == 4) ? a4 : r;                    \
+      r = (i == 5) ? a5 : r; return r;                                 \
+    }                                                                  \
+    DEV void set(int i, T x) {                                         \
+      a0 = (i == 0) ? x : a0; a1 = (i == 1) ? x : a1;                  \
+      a2 = (i == 2) ? x : a2; a3 = (i == 3) ? x : a3;                  \
+      a4 = (i == 4) ? x : a4; a5 = (i == 5) ? x : a5;                  \
+    }                                                                  \
+  };
+#define DN_REGSEL8(T, NAME)                                            \
+  struct NAME {                                                        \
+    T a0, a1, a2, a3, a4, a5, a6, a7;                                  \
+    DEV T get(int i) const {                                           \
+      T r = a0;                                                        \
+      r = (i == 1) ? a1 : r; r = (i == 2) ? a2 : r;                    \
+      r = (i == 3) ? a3 : r; r = (i == 4) ? a4 : r;                    \
+      r = (i == 5) ? a5 : r; r = (i == 6) ? a6 : r;                    \
+      r = (i == 7) ? a7 : r; return r;                                 \
+    }                                                                  \
+    DEV void set(int i, T x) {                                         \
+      a0 = (i == 0) ? x : a0; a1 = (i == 1) ? x : a1;                  \
+      a2 = (i == 2) ? x : a2; a3 = (i == 3) ? x : a3;                  \
+      a4 = (i == 4) ? x : a4; a5 = (i == 5) ? x : a5;                  \
+      a6 = (i == 6) ? x : a6; a7 = (i == 7) ? x : a7;                  \
+    }                                                                  \
+  };
+DN_REGSEL6(uint64_t, Sig6)
+DN_REGSEL8(uint64_t, U64x8)
+DN_REGSEL8(double, F64x8)
+DN_REGSEL8(uint32_t, U32x8)
 
 template <typename T>
 DEV T atomic_load_relaxed(const T* p) {
@@ -173,14 +221,11 @@ struct FV {
 
 // v * 10^ex without libm pow (exact for |ex| <= 22 when v < 2^53)
 DEV double scale10(double v, long ex) {
-  const double P10[] = {1e0,1e1,1e2,1e3,1e4,1e5,1e6,1e7,1e8,1e9,1e10,
-                        1e11,1e12,1e13,1e14,1e15,1e16,1e17,1e18,1e19,
-                        1e20,1e21,1e22};
   if (ex > 350) return v * __builtin_inf();
   if (ex < -350) return v * 0.0;
   while (ex > 22) { v *= 1e22; ex -= 22; }
   while (ex < -22) { v /= 1e22; ex += 22; }
-  return ex >= 0 ? v * P10[ex] : v / P10[-ex];
+  return ex >= 0 ? v * P10_TBL[ex] : v / P10_TBL[-ex];
 }
 
 struct NumOut { double v; bool ok; };
@@ -405,8 +450,7 @@ DEV DateOut parse_iso_ms(Bytes BV, uint32_t off, uint32_t len) {
   }
   if (p != j) return out;
   if (month < 1 || month > 12) return out;
-  const int dim_[12] = {31,28,31,30,31,30,31,31,30,31,30,31};
-  long dim = dim_[month-1] + ((month == 2 && is_leap(year)) ? 1 : 0);
+  long dim = DIM_TBL[month-1] + ((month == 2 && is_leap(year)) ? 1 : 0);
   if (day < 1 || day > dim) return out;
   if (hh > 24 || mm > 59 || ss > 59) return out;
   long days = days_from_civil(year, month, day);
@@ -427,8 +471,8 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
                       const PlanView& P, FV& fv, uint8_t& top_type) {
   Cursor c; c.init(BV, start, end);
 
-  uint64_t sig_stack[SIG_DEPTH];  // parent path sig per object depth
-  int8_t cont_slot[MAX_DEPTH];    // captured-container slot per depth
+  uint64_t sig_stack[SIG_DEPTH];  // parent sig per depth
+  uint64_t cont_slot = ~0ull;  // 5 bits/depth: captured slot or 31
   uint32_t is_arr_bits = 0;       // bit d: container at depth d is array
   int depth = 0;                  // container depth (0 = at top value)
   int arr_depth = 0;              // number of array containers on stack
@@ -525,8 +569,12 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
           // fall through to "after value"
         } else {
           // push object frame
-          cont_slot[depth] = (int8_t)((cur_capture && arr_depth == 0)
-                                          ? cur_slot : -1);
+          {
+            uint64_t cs = (cur_capture && arr_depth == 0)
+                              ? (uint64_t)cur_slot : 31ull;
+            cont_slot = (cont_slot & ~(31ull << (5 * depth)))
+                        | (cs << (5 * depth));
+          }
           if (depth < SIG_DEPTH) sig_stack[depth] = cur_sig;
           is_arr_bits &= ~(1u << depth);
           depth++;
@@ -554,8 +602,12 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
           voff = vstart; vlen = c.pos - vstart;
           is_container = false;
         } else {
-          cont_slot[depth] = (int8_t)((cur_capture && arr_depth == 0)
-                                          ? cur_slot : -1);
+          {
+            uint64_t cs = (cur_capture && arr_depth == 0)
+                              ? (uint64_t)cur_slot : 31ull;
+            cont_slot = (cont_slot & ~(31ull << (5 * depth)))
+                        | (cs << (5 * depth));
+          }
           if (depth < SIG_DEPTH) sig_stack[depth] = cur_sig;
           is_arr_bits |= (1u << depth);
           depth++;
@@ -610,9 +662,10 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
       if (b == ',') { expect_value = true; cur_capture = false; cur_slot = -1; continue; }
       if (b == ']') {
         depth--; arr_depth--;
-        if (cont_slot[depth] >= 0)
-          fv.set_len(cont_slot[depth],
-                     c.pos - fv.get_soff(cont_slot[depth]));
+        {
+          int cs = (int)((cont_slot >> (5 * depth)) & 31ull);
+          if (cs != 31) fv.set_len(cs, c.pos - fv.get_soff(cs));
+        }
         // restore parent sig (not needed for captures inside arrays)
         cur_sig = (depth < SIG_DEPTH) ? sig_stack[depth] : 0;
         continue;  // still "after value" for the parent
@@ -634,9 +687,10 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
       }
       if (b == '}') {
         depth--;
-        if (cont_slot[depth] >= 0)
-          fv.set_len(cont_slot[depth],
-                     c.pos - fv.get_soff(cont_slot[depth]));
+        {
+          int cs = (int)((cont_slot >> (5 * depth)) & 31ull);
+          if (cs != 31) fv.set_len(cs, c.pos - fv.get_soff(cs));
+        }
         cur_sig = (depth < SIG_DEPTH) ? sig_stack[depth] : 0;
         continue;
       }
@@ -727,17 +781,18 @@ DEV int eval_leaf(const PlanView& P, Bytes BV, const FV& fv,
 DEV int eval_predicate(const PlanView& P, Bytes BV,
                        const FV& fv, int prog_id) {
   int idx = P.prog_bounds[prog_id * 2 + 0];
-  struct Frame { int16_t op; int16_t remaining; int32_t end; };
-  Frame stk[PRED_STACK];
+  // frame packed into u64: op(1b) | remaining(23b) | end(32b), kept in
+  // registers via RegArr (a plain array would spill to scratch)
+  uint64_t stk[PRED_STACK];
   int sp = 0;
   int result;
   while (true) {
     int op = P.prog_nodes[idx * 4 + 0];
     if (op == OP_AND || op == OP_OR) {
       if (sp >= PRED_STACK) return -1;
-      stk[sp].op = (int16_t)op;
-      stk[sp].remaining = (int16_t)P.prog_nodes[idx * 4 + 1];
-      stk[sp].end = P.prog_nodes[idx * 4 + 3];
+      stk[sp] = ((uint64_t)(op == OP_OR)
+                  | ((uint64_t)P.prog_nodes[idx * 4 + 1] << 1)
+                  | ((uint64_t)P.prog_nodes[idx * 4 + 3] << 32));
       sp++;
       idx++;
       continue;
@@ -754,14 +809,17 @@ DEV int eval_predicate(const PlanView& P, Bytes BV,
     // unwind (short-circuit exactly like sequential evaluation)
     while (sp > 0) {
       if (result == -1) return -1;  // throw propagates
-      Frame& t = stk[sp - 1];
-      bool sc = (t.op == OP_AND && result == 0) ||
-                (t.op == OP_OR && result == 1);
-      t.remaining--;
-      if (sc || t.remaining == 0) {
-        idx = t.end;
+      uint64_t f = stk[sp - 1];
+      bool is_or = f & 1;
+      uint32_t remaining = (uint32_t)(f >> 1) & 0x7FFFFF;
+      bool sc = (!is_or && result == 0) || (is_or && result == 1);
+      remaining--;
+      if (sc || remaining == 0) {
+        idx = (int)(f >> 32);
         sp--;
       } else {
+        stk[sp - 1] = (f & 0xFFFFFFFF00000001ull)
+                      | ((uint64_t)remaining << 1);
         break;  // evaluate next child at idx
       }
     }
@@ -874,14 +932,18 @@ DEV bool agg_insert(const AggTable& T, const uint32_t* key, int nk,
       uint32_t st = atomic_load_relaxed(&T.state[s]);
       if (st == SLOT_READY) {
         bool same = true;
-        for (int k = 0; k < nk; k++)
-          if (atomic_load_relaxed(&T.keys[s * MAX_KEY + k]) != key[k]) { same = false; break; }
+#pragma unroll
+        for (int k = 0; k < MAX_KEY; k++)
+          if (k < nk &&
+              atomic_load_relaxed(&T.keys[s * MAX_KEY + k]) != key[k])
+            same = false;
         if (same) { atomicAdd(&T.count[s], w); return true; }
         break;
       }
       if (st == SLOT_EMPTY) {
         uint32_t prev = atomicCAS(&T.state[s], SLOT_EMPTY, SLOT_CLAIMED);
         if (prev == SLOT_EMPTY) {
+#pragma unroll
           for (int k = 0; k < MAX_KEY; k++)
             atomic_store_relaxed(&T.keys[s * MAX_KEY + k],
                                  k < nk ? key[k] : 0u);
@@ -1055,7 +1117,7 @@ __global__ void scan_kernel(ScanArgs A) {
   fv.tid = threadIdx.x;
 
   double synth_val[MAX_SYNTH];
-  uint8_t synth_ok[MAX_SYNTH];
+  uint32_t synth_ok = 0;  // 2 bits per synthetic field
 
   uint32_t nlines = *A.nlines_ptr;
   if (nlines > A.pos_cap) nlines = A.pos_cap;
@@ -1120,21 +1182,24 @@ __global__ void scan_kernel(ScanArgs A) {
 
         if (keep == 1) {
           // synthetic date fields (shared across metrics)
+          synth_ok = 0;
           for (int si = 0; si < P.ns; si++) {
             int slot = P.synth_slots[si];
             uint8_t t = fv.get_type(slot);
-            if (t == T_MISSING) { synth_ok[si] = 2; continue; }   // undef
-            if (t == T_NUM) { synth_ok[si] = 1; synth_val[si] = fv.get_num(slot); continue; }
-            if (t == T_STR) {
-              DateOut d = parse_iso_ms(BV, fv.get_soff(slot), fv.get_slen(slot));
+            uint32_t ok;
+            if (t == T_MISSING) ok = 2;                           // undef
+            else if (t == T_NUM) {
+              ok = 1; synth_val[si] = fv.get_num(slot);
+            } else if (t == T_STR) {
+              DateOut d = parse_iso_ms(BV, fv.get_soff(slot),
+                                       fv.get_slen(slot));
               if (d.ok) {
                 long long secs = d.ms >= 0 ? d.ms / 1000
-                                           : (d.ms - 999) / 1000;  // floor
-                synth_ok[si] = 1; synth_val[si] = (double)secs;
-              } else synth_ok[si] = 3;                             // baddate
-              continue;
-            }
-            synth_ok[si] = 3;  // bool/null/obj/arr: Date.parse fails
+                                           : (d.ms - 999) / 1000;
+                ok = 1; synth_val[si] = (double)secs;
+              } else ok = 3;                                      // baddate
+            } else ok = 3;  // bool/null/obj/arr: Date.parse fails
+            synth_ok |= ok << (2 * si);
           }
 
           // per-metric pipeline
@@ -1152,8 +1217,9 @@ __global__ void scan_kernel(ScanArgs A) {
             bool sok = true;
             for (int k = 0; k < M[3]; k++) {
               int si = P.synth_req[M[4] + k];
-              if (synth_ok[si] != 1) {
-                atomicAdd(&mc[synth_ok[si] == 2 ? CM_UNDEF : CM_BADDATE], 1ull);
+              uint32_t ok = (synth_ok >> (2 * si)) & 3u;
+              if (ok != 1) {
+                atomicAdd(&mc[ok == 2 ? CM_UNDEF : CM_BADDATE], 1ull);
                 sok = false; break;
               }
             }
@@ -1180,8 +1246,9 @@ __global__ void scan_kernel(ScanArgs A) {
               uint8_t t; double num = 0.0; uint32_t so = 0, sl = 0;
               if (B[0] == 1) {  // synthetic date value
                 int si = B[1];
-                if (synth_ok[si] == 1) { t = T_NUM; num = synth_val[si]; }
-                else t = T_MISSING;   // cannot happen if required above
+                if (((synth_ok >> (2 * si)) & 3u) == 1) {
+                  t = T_NUM; num = synth_val[si];
+                } else t = T_MISSING;  // cannot happen: required above
               } else {
                 int slot = B[1];
                 t = fv.get_type(slot);
@@ -1236,11 +1303,16 @@ __global__ void scan_kernel(ScanArgs A) {
                 if (id == 0xFFFFFFFFu) { overflow = true; break; }
                 code = make_code(TAG_STR, id);
               }
-              key[bi] = code;
+              // constant-index write keeps key[] in registers
+#pragma unroll
+              for (int kk = 0; kk < MAX_KEY; kk++)
+                if (kk == bi) key[kk] = code;
             }
             if (overflow) { atomicAdd(&lcnt[C_OVERFLOW], 1ull); continue; }
             if (drop) { atomicAdd(&mc[CM_NONNUMERIC], 1ull); continue; }
-            for (int k = nk; k < MAX_KEY; k++) key[k] = 0;
+#pragma unroll
+            for (int k = 0; k < MAX_KEY; k++)
+              if (k >= nk) key[k] = 0;
 
             // LDS combining cache: hash (metric, key)
             uint64_t kh = mix64((uint64_t)m * 0x9E3779B97F4A7C15ull + 1);

@@ -220,23 +220,7 @@ class _ScanContext(object):
         self._copy_ev.record()
         self.eng.ops.newline_index(dev_data, 0, n, self._segs,
                                    self._pos, self._nlines)
-        self.eng.ops.scan_chunk(
-            dev_data, self._pos, self._nlines, 0,
-            self.field_sigs, self.prog_nodes, self.prog_bounds,
-            self.const_meta, self.const_dvals, self.const_bytes,
-            self.synth_slots, self.cplan.n_synth,
-            self.metric_rows, self.synth_req,
-            self.bd_rows, self.bd_steps,
-            self.cplan.value_slot,
-            getattr(self.cplan, "fields_slot", -1),
-            self.cplan.data_format == "json-skinner",
-            self.table_descs,
-            self.sd["state"], self.sd["hash"], self.sd["id"],
-            self.sd["off"], self.sd["len"], self.sd["data"],
-            self.sd["used"], self.sd["next"],
-            self.nd["state"], self.nd["bits"], self.nd["id"],
-            self.nd["next"],
-            self.counters)
+        self._scan_call(dev_data, 0)
 
     def overflowed(self):
         return int(self.counters[5].item()) > 0

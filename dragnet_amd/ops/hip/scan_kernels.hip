@@ -1153,7 +1153,10 @@ __global__ void scan_kernel(ScanArgs A) {
 
     if (active) {
       atomicAdd(&lcnt[C_LINES], 1ull);
-      for (int f = 0; f < nf; f++) fv.set(f, T_MISSING, 0, 0, 0.0);
+      // only the type lane needs initializing: soff/slen/num are read
+      // only after a capture set them
+      for (int f = 0; f < nf; f++)
+        fv.type[f * BLOCK + fv.tid] = T_MISSING;
 
       uint32_t start = r ? A.nl_pos[r - 1] + 1 : A.first_start;
       uint32_t end = A.nl_pos[r];

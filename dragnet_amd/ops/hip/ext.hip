@@ -6,6 +6,7 @@
 // the Python engine (dragnet_amd/engine/gpu.py).
 
 #include <torch/extension.h>
+#include <cstdlib>
 #include <c10/hip/HIPStream.h>
 
 #include "scan_kernels.hip"
@@ -173,8 +174,20 @@ void scan_chunk(
   uint32_t blocks = (A.pos_cap + BLOCK - 1) / BLOCK;
   if (blocks > 2048) blocks = 2048;
 
-  hipLaunchKernelGGL(scan_kernel, dim3(blocks), dim3(BLOCK), lds,
-                     current_stream(), A);
+  const char* mw_env = getenv("DRAGNET_MIN_WAVES");
+  int mw = mw_env ? atoi(mw_env) : 0;
+  if (mw == 2)
+    hipLaunchKernelGGL(scan_kernel_mw<2>, dim3(blocks), dim3(BLOCK),
+                       lds, current_stream(), A);
+  else if (mw == 3)
+    hipLaunchKernelGGL(scan_kernel_mw<3>, dim3(blocks), dim3(BLOCK),
+                       lds, current_stream(), A);
+  else if (mw == 4)
+    hipLaunchKernelGGL(scan_kernel_mw<4>, dim3(blocks), dim3(BLOCK),
+                       lds, current_stream(), A);
+  else
+    hipLaunchKernelGGL(scan_kernel, dim3(blocks), dim3(BLOCK), lds,
+                       current_stream(), A);
   hipError_t err = hipGetLastError();
   TORCH_CHECK(err == hipSuccess, "scan_kernel launch failed: ",
               hipGetErrorString(err));

@@ -1077,9 +1077,27 @@ __global__ void newline_write_kernel(const uint8_t* data,
   }
 }
 
+DEV void scan_kernel_body(char* smem, ScanArgs A);
+
 __launch_bounds__(BLOCK)
 __global__ void scan_kernel(ScanArgs A) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
+  extern __shared__ __attribute__((aligned(16))) char smem0[];
+  scan_kernel_body(smem0, A);
+}
+
+// occupancy-capped variants (min waves per SIMD; constrains the
+// register allocator) — selected at runtime for A/B measurement
+template <int MW>
+__launch_bounds__(BLOCK, MW)
+__global__ void scan_kernel_mw(ScanArgs A) {
+  extern __shared__ __attribute__((aligned(16))) char smem1[];
+  scan_kernel_body(smem1, A);
+}
+template __global__ void scan_kernel_mw<2>(ScanArgs);
+template __global__ void scan_kernel_mw<3>(ScanArgs);
+template __global__ void scan_kernel_mw<4>(ScanArgs);
+
+DEV void scan_kernel_body(char* smem, ScanArgs A) {
   const PlanView& P = A.P;
   const int nf = P.nf;
 

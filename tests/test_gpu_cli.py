@@ -102,3 +102,17 @@ def test_cli_counters_gpu_vs_cpu(dn, fixture_tree, monkeypatch):
     # the pipeline counter dumps agree stage by stage (FindX stages and
     # json parser/Datetime/Aggregator taxonomies are engine-agnostic)
     assert g.err == c.err
+
+
+def test_cli_sharded_backend_gpu(dn, fixture_tree, monkeypatch):
+    """Sharded backend, world=1, GPU engine: equals the file backend."""
+    monkeypatch.setenv("DRAGNET_ENGINE", "gpu")
+    r = dn("datasource-add", "sh", "--backend=sharded",
+           "--path=" + fixture_tree)
+    assert r.code == 0, r.err
+    r = dn("datasource-add", "fl", "--path=" + fixture_tree)
+    assert r.code == 0, r.err
+    a = dn("scan", "-b", "req.method", "sh")
+    b = dn("scan", "-b", "req.method", "fl")
+    assert a.code == 0 and b.code == 0, (a.err, b.err)
+    assert a.out == b.out

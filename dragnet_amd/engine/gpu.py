@@ -33,6 +33,12 @@ CM_N = 8
 MAX_KEY = 8
 
 
+def _pad(n):
+    """Pad a chunk length so the cursor's 8-byte windows never read
+    past the buffer (>= 8 bytes of newline slack, 16B aligned)."""
+    return (n + 8 + 15) & ~15
+
+
 def _env_int(name, default):
     try:
         return int(os.environ.get(name, default))
@@ -203,7 +209,7 @@ class _ScanContext(object):
         n = len(buf)
         if n == 0:
             return
-        padded = (n + 15) & ~15
+        padded = _pad(n)
         self._ensure_buffers(padded)
         pin = self._pinned
         # the previous chunk's async H2D copy must complete before the
@@ -235,7 +241,7 @@ class _ScanContext(object):
         slice k (copy stream + events)."""
         torch = self.t
         n = len(buf)
-        padded = (n + 15) & ~15
+        padded = _pad(n)
         self._ensure_buffers(padded)
         pin = self._pinned
         pin[:n] = torch.frombuffer(bytearray(buf), dtype=torch.uint8)

@@ -175,8 +175,10 @@ void scan_chunk(
   uint32_t blocks = (A.pos_cap + BLOCK - 1) / BLOCK;
   if (blocks > 2048) blocks = 2048;
 
+  // default 4 waves/SIMD: the parse kernel is latency-bound and the
+  // register-allocator cap measured +17% over the unconstrained build
   const char* mw_env = getenv("DRAGNET_MIN_WAVES");
-  int mw = mw_env ? atoi(mw_env) : 0;
+  int mw = mw_env ? atoi(mw_env) : 4;
   if (mw == 2)
     hipLaunchKernelGGL(scan_kernel_mw<2>, dim3(blocks), dim3(BLOCK),
                        lds, current_stream(), A);
@@ -185,6 +187,12 @@ void scan_chunk(
                        lds, current_stream(), A);
   else if (mw == 4)
     hipLaunchKernelGGL(scan_kernel_mw<4>, dim3(blocks), dim3(BLOCK),
+                       lds, current_stream(), A);
+  else if (mw == 5)
+    hipLaunchKernelGGL(scan_kernel_mw<5>, dim3(blocks), dim3(BLOCK),
+                       lds, current_stream(), A);
+  else if (mw == 6)
+    hipLaunchKernelGGL(scan_kernel_mw<6>, dim3(blocks), dim3(BLOCK),
                        lds, current_stream(), A);
   else
     hipLaunchKernelGGL(scan_kernel, dim3(blocks), dim3(BLOCK), lds,

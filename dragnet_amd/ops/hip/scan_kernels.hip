@@ -118,17 +118,31 @@ struct Bytes {
 struct Cursor {
   Bytes B;
   uint32_t pos, end;
+  uint64_t win;     // 8 record bytes at [wbase, wbase+8)
+  uint32_t wbase;
 
   DEV void init(Bytes b, uint32_t p, uint32_t e) {
     B = b; pos = p; end = e;
+    wbase = p;
+    __builtin_memcpy(&win, B.ptr(p), 8);
   }
-  DEV uint8_t byte_at(uint32_t p) const { return B.at(p); }
+  // NOTE: refill may read up to 7 bytes past `end`; the host pads
+  // every chunk with >= 8 newline bytes (engine/gpu.py _pad()).
+  DEV uint8_t byte_at(uint32_t p) {
+    uint32_t d = p - wbase;
+    if (d >= 8u) {
+      wbase = p;
+      __builtin_memcpy(&win, B.ptr(p), 8);
+      d = 0;
+    }
+    return (uint8_t)(win >> (8u * d));
+  }
   DEV bool eof() const { return pos >= end; }
-  DEV uint8_t peek() const { return B.at(pos); }
-  DEV uint8_t next() { return B.at(pos++); }
+  DEV uint8_t peek() { return byte_at(pos); }
+  DEV uint8_t next() { return byte_at(pos++); }
   DEV void skip_ws() {
     while (pos < end) {
-      uint8_t b = B.at(pos);
+      uint8_t b = byte_at(pos);
       if (b == ' ' || b == '\t' || b == '\r' || b == '\n') pos++;
       else break;
     }
@@ -1096,6 +1110,8 @@ __global__ void scan_kernel_mw(ScanArgs A) {
 template __global__ void scan_kernel_mw<2>(ScanArgs);
 template __global__ void scan_kernel_mw<3>(ScanArgs);
 template __global__ void scan_kernel_mw<4>(ScanArgs);
+template __global__ void scan_kernel_mw<5>(ScanArgs);
+template __global__ void scan_kernel_mw<6>(ScanArgs);
 
 DEV void scan_kernel_body(char* smem, ScanArgs A) {
   const PlanView& P = A.P;

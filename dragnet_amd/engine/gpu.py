@@ -77,14 +77,15 @@ class GpuEngine(object):
         for attempt in range(3):
             ctx = _ScanContext(self, cplan, agg_slots, dict_slots,
                                dict_data)
-            src = byte_source if byte_source is not None \
-                else _read_files(files, self.chunk_bytes)
-            if byte_source is not None and attempt > 0:
-                raise RuntimeError(
-                    "table overflow on non-restartable byte source")
-            for chunk in src:
-                ctx.feed(chunk)
-            ctx.flush()
+            if byte_source is not None:
+                if attempt > 0:
+                    raise RuntimeError(
+                        "table overflow on non-restartable byte source")
+                for chunk in byte_source:
+                    ctx.feed(chunk)
+                ctx.flush()
+            else:
+                ctx.scan_files(files)
             if ctx.overflowed():
                 agg_slots *= 8
                 dict_slots *= 8
@@ -230,6 +231,97 @@ class _ScanContext(object):
 
     def overflowed(self):
         return int(self.counters[5].item()) > 0
+
+    # ---- zero-copy file scanning ----
+
+    def scan_files(self, files):
+        """Scan the concatenated bytes of `files` reading directly
+        into ping-pong pinned buffers (no intermediate Python bytes):
+        each chunk is cut at its last newline and the tail is carried
+        into the head of the other buffer while the GPU works on the
+        previous chunk."""
+        torch = self.t
+        cap = self.eng.chunk_bytes
+        if self._pinned is None:
+            self._ensure_buffers(_pad(cap))
+        pins = [self._pinned,
+                torch.empty(self._pinned.numel(), dtype=torch.uint8,
+                            pin_memory=True)]
+        views = [memoryview(p.numpy()) for p in pins]
+        evs = [None, None]
+
+        def last_newline(buf, n):
+            probe = max(0, n - (1 << 16))
+            cut = bytes(buf[probe:n]).rfind(b"\n")
+            if cut >= 0:
+                return probe + cut
+            return bytes(buf[:probe]).rfind(b"\n")
+
+        cur = 0
+        head = 0  # carried tail bytes at the front of pins[cur]
+        fiter = iter(files)
+        f = None
+        while True:
+            # pins[cur] may still be the source of an in-flight H2D
+            if evs[cur] is not None:
+                evs[cur].synchronize()
+                evs[cur] = None
+            n = head
+            eof = False
+            while n < cap:
+                if f is None:
+                    try:
+                        f = open(next(fiter), "rb", buffering=0)
+                    except StopIteration:
+                        eof = True
+                        break
+                got = f.readinto(views[cur][n:cap])
+                if not got:
+                    f.close()
+                    f = None
+                    continue
+                n += got
+            if n == 0:
+                break
+            cut = last_newline(views[cur], n)
+            if eof and cut < n - 1:
+                # final partial line: terminate it
+                views[cur][n:n + 1] = b"\n"
+                n += 1
+                cut = n - 1
+            if cut < 0:
+                if n >= cap:
+                    raise RuntimeError(
+                        "record exceeds the chunk buffer (%d bytes); "
+                        "raise DRAGNET_CHUNK_MB" % cap)
+                tail = n  # no newline yet: carry the whole buffer
+            else:
+                evs[cur] = self._run_pinned(pins[cur], cut + 1)
+                tail = n - (cut + 1)
+            nxt = 1 - cur
+            if tail:
+                if evs[nxt] is not None:
+                    evs[nxt].synchronize()
+                    evs[nxt] = None
+                views[nxt][:tail] = views[cur][cut + 1:n]
+            head = tail
+            cur = nxt
+            if eof:
+                break
+
+    def _run_pinned(self, pin, n):
+        """H2D + kernels for pin[:n] (already newline-terminated)."""
+        torch = self.t
+        padded = _pad(n)
+        pin[n:padded] = 10
+        dev_data = self._dev_data
+        dev_data[:padded].copy_(pin[:padded], non_blocking=True)
+        ev = torch.cuda.Event()
+        ev.record()
+        self.eng.ops.newline_index(dev_data, 0, n, self._segs,
+                                   self._pos, self._nlines)
+        self._scan_call(dev_data, 0)
+        return ev
 
     # ---- resident-pool path (bench / repeated scans) ----
 

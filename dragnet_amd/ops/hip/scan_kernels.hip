@@ -564,7 +564,6 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
       uint8_t vtype = T_MISSING;
       uint32_t voff = 0, vlen = 0;
       double vnum = 0.0;
-      bool is_container = false;
 
       if (b == '{') {
         uint32_t vstart = c.pos;
@@ -579,7 +578,6 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
           c.pos++;
           vtype = T_OBJ;  // empty object: treat as closed value
           voff = vstart; vlen = c.pos - vstart;
-          is_container = false;
           // fall through to "after value"
         } else {
           // push object frame
@@ -614,7 +612,6 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
           c.pos++;
           vtype = T_ARR;
           voff = vstart; vlen = c.pos - vstart;
-          is_container = false;
         } else {
           {
             uint64_t cs = (cur_capture && arr_depth == 0)

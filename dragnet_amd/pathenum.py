@@ -129,3 +129,28 @@ def enumerate_paths(pattern, start_ms, end_ms):
         if t_ms >= end_ms:
             break
     return out
+
+
+def main():
+    """CLI harness (the reference tools/pathenum analog):
+    pathenum PATTERN START END"""
+    import sys
+
+    from . import jsdate
+    if len(sys.argv) != 4:
+        sys.stderr.write("usage: pathenum PATTERN START END\n")
+        return 2
+    try:
+        for p in enumerate_paths(sys.argv[1],
+                                 jsdate.parse_ms(sys.argv[2]),
+                                 jsdate.parse_ms(sys.argv[3])):
+            print(p)
+    except PathEnumError as e:
+        sys.stderr.write("pathenum: %s\n" % e)
+        return 1
+    return 0
+
+
+if __name__ == "__main__":
+    import sys
+    sys.exit(main())

@@ -1354,7 +1354,10 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
             if (kh == 0) kh = 1;
             bool cached = false;
             uint32_t ci = (uint32_t)kh & (LDS_CACHE - 1);
-            for (int attempt = 0; attempt < 2; attempt++) {
+            // 8 probes: a hot key that loses every probe degrades to
+            // per-record contended global inserts (measured 4x slower
+            // on low-cardinality ordinal breakdowns with 2 probes)
+            for (int attempt = 0; attempt < 8; attempt++) {
               unsigned long long prev = atomicCAS(
                   (unsigned long long*)&cache[ci].hash, 0ull,
                   (unsigned long long)kh);

@@ -1352,6 +1352,30 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
             uint64_t kh = mix64((uint64_t)m * 0x9E3779B97F4A7C15ull + 1);
             for (int k = 0; k < MAX_KEY; k++) kh = mix64(kh ^ key[k]);
             if (kh == 0) kh = 1;
+
+            // Wave-level pre-combining: lanes carrying the same key
+            // elect one leader per distinct kh which adds the whole
+            // group's weight — one LDS atomic per distinct key per
+            // wavefront instead of one per record.  (Only when every
+            // weight is 1: json format; skinner weights vary.)
+            if (!A.data_format_skinner) {
+              uint64_t unproc = __ballot(true);  // lanes still here
+              bool leader = false;
+              double wsum = 0.0;
+              while (unproc) {
+                int src = (int)(__ffsll((long long)unproc) - 1);
+                uint64_t src_kh = __shfl(kh, src);
+                bool same = (kh == src_kh);
+                uint64_t grp = __ballot(same);
+                if (same && __lane_id() == src) {
+                  leader = true;
+                  wsum = (double)__popcll(grp);
+                }
+                unproc &= ~grp;
+              }
+              if (!leader) continue;  // combined into the leader's add
+              weight = wsum;
+            }
             bool cached = false;
             uint32_t ci = (uint32_t)kh & (LDS_CACHE - 1);
             // 8 probes: a hot key that loses every probe degrades to

@@ -482,10 +482,12 @@ DEV DateOut parse_iso_ms(Bytes BV, uint32_t off, uint32_t len) {
 // preinitialized to T_MISSING by the caller).  Returns false on invalid
 // JSON.  top_type receives the top-level value type.
 DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
-                      const PlanView& P, FV& fv, uint8_t& top_type) {
+                      const PlanView& P, FV& fv, uint8_t& top_type,
+                      uint64_t* sig_lds) {
   Cursor c; c.init(BV, start, end);
+  // parent-signature stack in LDS: [depth * BLOCK + tid]
+  #define sig_stack_at(d) sig_lds[(d) * BLOCK + (uint32_t)threadIdx.x]
 
-  uint64_t sig_stack[SIG_DEPTH];  // parent sig per depth
   uint64_t cont_slot = ~0ull;  // 5 bits/depth: captured slot or 31
   uint32_t is_arr_bits = 0;       // bit d: container at depth d is array
   int depth = 0;                  // container depth (0 = at top value)
@@ -587,7 +589,7 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
             cont_slot = (cont_slot & ~(31ull << (5 * depth)))
                         | (cs << (5 * depth));
           }
-          if (depth < SIG_DEPTH) sig_stack[depth] = cur_sig;
+          if (depth < SIG_DEPTH) sig_stack_at(depth) = cur_sig;
           is_arr_bits &= ~(1u << depth);
           depth++;
           // parse first key
@@ -619,7 +621,7 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
             cont_slot = (cont_slot & ~(31ull << (5 * depth)))
                         | (cs << (5 * depth));
           }
-          if (depth < SIG_DEPTH) sig_stack[depth] = cur_sig;
+          if (depth < SIG_DEPTH) sig_stack_at(depth) = cur_sig;
           is_arr_bits |= (1u << depth);
           depth++;
           arr_depth++;
@@ -678,14 +680,14 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
           if (cs != 31) fv.set_len(cs, c.pos - fv.get_soff(cs));
         }
         // restore parent sig (not needed for captures inside arrays)
-        cur_sig = (depth < SIG_DEPTH) ? sig_stack[depth] : 0;
+        cur_sig = (depth < SIG_DEPTH) ? sig_stack_at(depth) : 0;
         continue;  // still "after value" for the parent
       }
       return false;
     } else {
       if (b == ',') {
         c.skip_ws();
-        uint64_t parent = (depth - 1 < SIG_DEPTH) ? sig_stack[depth - 1] : 0;
+        uint64_t parent = (depth - 1 < SIG_DEPTH) ? sig_stack_at(depth - 1) : 0;
         uint64_t ksig;
         if (!parse_key(parent, depth == 1, ksig)) return false;
         c.skip_ws();
@@ -702,7 +704,7 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
           int cs = (int)((cont_slot >> (5 * depth)) & 31ull);
           if (cs != 31) fv.set_len(cs, c.pos - fv.get_soff(cs));
         }
-        cur_sig = (depth < SIG_DEPTH) ? sig_stack[depth] : 0;
+        cur_sig = (depth < SIG_DEPTH) ? sig_stack_at(depth) : 0;
         continue;
       }
       return false;
@@ -1130,6 +1132,10 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
   unsigned long long* lcnt = reinterpret_cast<unsigned long long*>(smem + off);
   const int NCNT = C_GLOBAL_N + P.nm * CM_N;
   off += (size_t)NCNT * sizeof(unsigned long long);
+  double* synth_lds = reinterpret_cast<double*>(smem + off);
+  off += (size_t)MAX_SYNTH * BLOCK * sizeof(double);
+  uint64_t* sig_lds = reinterpret_cast<uint64_t*>(smem + off);
+  off += (size_t)SIG_DEPTH * BLOCK * sizeof(uint64_t);
   off = (off + 15) & ~(size_t)15;
   uint8_t* tile = reinterpret_cast<uint8_t*>(smem + off);  // staging
 
@@ -1147,7 +1153,8 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
   fv.type = fv_type; fv.soff = fv_soff; fv.slen = fv_slen; fv.num = fv_num;
   fv.tid = threadIdx.x;
 
-  double synth_val[MAX_SYNTH];
+  // synthetic values live in LDS (keeps the MW=4 build spill-free)
+  #define synth_val_at(si) synth_lds[(si) * BLOCK + threadIdx.x]
   uint32_t synth_ok = 0;  // 2 bits per synthetic field
 
   uint32_t nlines = *A.nlines_ptr;
@@ -1193,7 +1200,7 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
       uint32_t end = A.nl_pos[r];
       uint8_t top_type;
       bool ok = (end > start) &&
-                parse_record(BV, start, end, P, fv, top_type);
+                parse_record(BV, start, end, P, fv, top_type, sig_lds);
       double weight = 1.0;
       if (ok && A.data_format_skinner) {
         // require: object top, a "fields" member, numeric "value"
@@ -1223,14 +1230,14 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
             uint32_t ok;
             if (t == T_MISSING) ok = 2;                           // undef
             else if (t == T_NUM) {
-              ok = 1; synth_val[si] = fv.get_num(slot);
+              ok = 1; synth_val_at(si) = fv.get_num(slot);
             } else if (t == T_STR) {
               DateOut d = parse_iso_ms(BV, fv.get_soff(slot),
                                        fv.get_slen(slot));
               if (d.ok) {
                 long long secs = d.ms >= 0 ? d.ms / 1000
                                            : (d.ms - 999) / 1000;
-                ok = 1; synth_val[si] = (double)secs;
+                ok = 1; synth_val_at(si) = (double)secs;
               } else ok = 3;                                      // baddate
             } else ok = 3;  // bool/null/obj/arr: Date.parse fails
             synth_ok |= ok << (2 * si);
@@ -1262,7 +1269,7 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
             // time filter on dn_ts (= last synth req when present)
             if (M[5]) {
               int si = P.synth_req[M[4] + M[3] - 1];
-              double ts = synth_val[si];
+              double ts = synth_val_at(si);
               if (!(ts >= (double)M[6] && ts < (double)M[7])) {
                 atomicAdd(&mc[CM_TIME_OUT], 1ull); continue;
               }
@@ -1281,7 +1288,7 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
               if (B[0] == 1) {  // synthetic date value
                 int si = B[1];
                 if (((synth_ok >> (2 * si)) & 3u) == 1) {
-                  t = T_NUM; num = synth_val[si];
+                  t = T_NUM; num = synth_val_at(si);
                 } else t = T_MISSING;  // cannot happen: required above
               } else {
                 int slot = B[1];

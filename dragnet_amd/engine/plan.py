@@ -94,7 +94,7 @@ class FieldTable(object):
 
 class ConstPool(object):
     def __init__(self):
-        self.metas = []   # (kind, str_off, str_len, dvalid)
+        self.metas = []   # (kind, str_off, str_len, dvalid, off2, len2)
         self.dvals = []
         self.bytes = bytearray()
 
@@ -109,13 +109,25 @@ class ConstPool(object):
             self.metas.append((CONST_NUM, 0, 0, 1))
             self.dvals.append(float(v))
         else:
+            import json as _json
             b = v.encode("utf-8")
             off = len(self.bytes)
             self.bytes.extend(b)
             num = krill.to_number(v)
             valid = 0 if num != num else 1
-            self.metas.append((CONST_STR, off, len(b), valid))
+            # records may carry the string in ESCAPED form; store the
+            # canonical JSON escaping as an alternate match target
+            esc = _json.dumps(v, ensure_ascii=False)[1:-1].encode("utf-8")
+            if esc != b:
+                off2 = len(self.bytes)
+                self.bytes.extend(esc)
+                alt = (off2, len(esc))
+            else:
+                alt = (0, 0)  # len2 == 0: no alternate form
+            self.metas.append((CONST_STR, off, len(b), valid) + alt)
             self.dvals.append(num if valid else 0.0)
+            return len(self.metas) - 1
+        self.metas.append(self.metas.pop() + (0, 0))
         return len(self.metas) - 1
 
 
@@ -281,7 +293,7 @@ def compile_plan(queries, ds_filter=None, time_field=None,
 
     programs = np.array(prog_nodes, dtype=np.int32).reshape(-1, 4)
     bounds = np.array(prog_bounds, dtype=np.int32).reshape(-1, 2)
-    const_meta = np.array(consts.metas, dtype=np.int32).reshape(-1, 4)
+    const_meta = np.array(consts.metas, dtype=np.int32).reshape(-1, 6)
     const_dvals = np.array(consts.dvals, dtype=np.float64)
     const_bytes = np.frombuffer(
         bytes(consts.bytes) or b"\0", dtype=np.uint8).copy()

@@ -113,7 +113,7 @@ struct PlanView {
   int nf;
   const int32_t* prog_nodes;    // [n_nodes][4]
   const int32_t* prog_bounds;   // [n_progs][2]
-  const int32_t* const_meta;    // [nc][4]  kind, off, len, dvalid
+  const int32_t* const_meta;    // [nc][6] kind, off, len, dvalid, off2, len2
   const double*  const_dvals;   // [nc]
   const uint8_t* const_bytes;
   const int32_t* synth_slots;   // [ns] source field slot

@@ -721,10 +721,10 @@ DEV int eval_leaf(const PlanView& P, Bytes BV, const FV& fv,
   uint8_t ft = fv.get_type(slot);
   if (ft == T_MISSING) return -1;  // krill: missing field -> throw
 
-  int ckind = P.const_meta[cidx * 4 + 0];
-  uint32_t coff = (uint32_t)P.const_meta[cidx * 4 + 1];
-  uint32_t clen = (uint32_t)P.const_meta[cidx * 4 + 2];
-  int cdvalid = P.const_meta[cidx * 4 + 3];
+  int ckind = P.const_meta[cidx * 6 + 0];
+  uint32_t coff = (uint32_t)P.const_meta[cidx * 6 + 1];
+  uint32_t clen = (uint32_t)P.const_meta[cidx * 6 + 2];
+  int cdvalid = P.const_meta[cidx * 6 + 3];
   double cdval = P.const_dvals[cidx];
 
   if (op == OP_EQ || op == OP_NE) {
@@ -744,6 +744,17 @@ DEV int eval_leaf(const PlanView& P, Bytes BV, const FV& fv,
           eq = true;
           for (uint32_t k = 0; k < fl; k++)
             if (data[fo + k] != P.const_bytes[coff + k]) { eq = false; break; }
+        }
+        // records may carry the constant in JSON-ESCAPED form: try
+        // the canonical escaped rendering too (plan.py ConstPool)
+        if (!eq) {
+          uint32_t co2 = (uint32_t)P.const_meta[cidx * 6 + 4];
+          uint32_t cl2 = (uint32_t)P.const_meta[cidx * 6 + 5];
+          if (cl2 != 0 && fl == cl2) {
+            eq = true;
+            for (uint32_t k = 0; k < fl; k++)
+              if (data[fo + k] != P.const_bytes[co2 + k]) { eq = false; break; }
+          }
         }
       } else {  // string vs number: ToNumber(field)
         double fn = js_to_number(BV, fv.get_soff(slot), fv.get_slen(slot));

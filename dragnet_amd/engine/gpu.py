@@ -325,13 +325,15 @@ class _ScanContext(object):
 
     # ---- resident-pool path (bench / repeated scans) ----
 
-    def stage_resident(self, buf, n_slices=6):
+    def stage_resident(self, buf, n_slices=None):
         """Stage a byte pool into the pinned buffer once; later
         scan_resident() calls re-run the H2D + kernels without the
         host-side copy.  The pool is split into newline-aligned
         slices so the H2D copy of slice k+1 overlaps the scan of
         slice k (copy stream + events)."""
         torch = self.t
+        if n_slices is None:
+            n_slices = _env_int("DRAGNET_SLICES", 6)
         n = len(buf)
         padded = _pad(n)
         self._ensure_buffers(padded)

@@ -61,3 +61,25 @@ for name, q in CASES2:
     torch.cuda.synchronize()
     dt = (time.time() - t0) / 6
     print("%-38s %7.2f GB/s  %6.1f M rec/s" % (name, len(pool)/dt/1e9, nrec/dt/1e6))
+
+# raw H2D link rate + finalize cost
+pin = torch.empty(256 << 20, dtype=torch.uint8, pin_memory=True)
+dev = torch.empty(256 << 20, dtype=torch.uint8, device="cuda")
+for _ in range(3):
+    dev.copy_(pin, non_blocking=True)
+torch.cuda.synchronize()
+t0 = time.time()
+for _ in range(10):
+    dev.copy_(pin, non_blocking=True)
+torch.cuda.synchronize()
+print("raw H2D: %.2f GB/s" % ((256 << 20) * 10 / (time.time() - t0) / 1e9))
+
+q = CASES[1][1]
+cplan = planmod.compile_plan([q])
+ctx = _ScanContext(eng, cplan, 1 << 18, 1 << 18, 32 << 20)
+ctx.stage_resident(pool)
+ctx.reset(); ctx.scan_resident(h2d=False); ctx.finalize([q])
+t0 = time.time()
+for _ in range(10):
+    ctx.finalize([q])
+print("finalize: %.2f ms" % ((time.time() - t0) / 10 * 1000))

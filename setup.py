@@ -13,6 +13,16 @@ from setuptools import setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
+# torch's hipify step copies ext.hip -> ext_hip.hip at first build and
+# ninja's depfile then tracks the COPY, so edits to the included kernel
+# sources do not trigger recompilation.  Force a clean kernel build.
+import glob
+import shutil
+
+for stale in glob.glob("dragnet_amd/ops/hip/*_hip.hip"):
+    os.unlink(stale)
+shutil.rmtree("build/temp.linux-x86_64-3.10", ignore_errors=True)
+
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
 
 setup(

@@ -129,6 +129,9 @@ struct Cursor {
   // NOTE: refill may read up to 7 bytes past `end`; the host pads
   // every chunk with >= 8 newline bytes (engine/gpu.py _pad()).
   DEV uint8_t byte_at(uint32_t p) {
+#ifdef DN_DIRECT_BYTES
+    return B.at(p);
+#else
     uint32_t d = p - wbase;
     if (d >= 8u) {
       wbase = p;
@@ -136,6 +139,7 @@ struct Cursor {
       d = 0;
     }
     return (uint8_t)(win >> (8u * d));
+#endif
   }
   DEV bool eof() const { return pos >= end; }
   DEV uint8_t peek() { return byte_at(pos); }

@@ -110,9 +110,18 @@ def main():
     if args.device_resident:
         ctx.scan_resident()  # prime HBM once
 
+    graph = None
+    if os.environ.get("DRAGNET_NO_GRAPH") != "1":
+        graph = ctx.make_graph(h2d=not args.device_resident)
+        log("hipGraph capture: %s"
+            % ("ok" if graph is not None else "unavailable"))
+
     def step():
-        ctx.reset()
-        ctx.scan_resident(h2d=not args.device_resident)
+        if graph is not None:
+            graph.replay()
+        else:
+            ctx.reset()
+            ctx.scan_resident(h2d=not args.device_resident)
         aggs, _stages = ctx.finalize([query])
         if dist is not None:
             merged = merge_tables_tensor(aggs[0], query, device)

@@ -36,7 +36,7 @@ MAX_KEY = 8
 def _pad(n):
     """Pad a chunk length so the cursor's 8-byte windows never read
     past the buffer (>= 8 bytes of newline slack, 16B aligned)."""
-    return (n + 8 + 15) & ~15
+    return (n + 16 + 15) & ~15
 
 
 def _env_int(name, default):

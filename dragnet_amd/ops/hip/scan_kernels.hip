@@ -118,27 +118,31 @@ struct Bytes {
 struct Cursor {
   Bytes B;
   uint32_t pos, end;
-  uint64_t win;     // 8 record bytes at [wbase, wbase+8)
+  uint64_t win, win2;  // 16 record bytes at [wbase, wbase+16)
   uint32_t wbase;
 
   DEV void init(Bytes b, uint32_t p, uint32_t e) {
     B = b; pos = p; end = e;
     wbase = p;
     __builtin_memcpy(&win, B.ptr(p), 8);
+    __builtin_memcpy(&win2, B.ptr(p) + 8, 8);
   }
-  // NOTE: refill may read up to 7 bytes past `end`; the host pads
-  // every chunk with >= 8 newline bytes (engine/gpu.py _pad()).
+  // NOTE: refill may read up to 15 bytes past `end`; the host pads
+  // every chunk with >= 8 newline bytes and 16B alignment slack
+  // (engine/gpu.py _pad()).
   DEV uint8_t byte_at(uint32_t p) {
 #ifdef DN_DIRECT_BYTES
     return B.at(p);
 #else
     uint32_t d = p - wbase;
-    if (d >= 8u) {
+    if (d >= 16u) {
       wbase = p;
       __builtin_memcpy(&win, B.ptr(p), 8);
+      __builtin_memcpy(&win2, B.ptr(p) + 8, 8);
       d = 0;
     }
-    return (uint8_t)(win >> (8u * d));
+    uint64_t w = d < 8u ? win : win2;
+    return (uint8_t)(w >> (8u * (d & 7u)));
 #endif
   }
   DEV bool eof() const { return pos >= end; }

@@ -1,0 +1,129 @@
+"""
+Differential fuzzing: random JSON records + random queries, GPU kernel
+vs CPU oracle (pytest -m gpu).
+
+Stays inside the documented parity envelope: canonical JSON escaping
+(json.dumps), nesting depth <= 6, numbers <= 15 significant digits.
+Everything else — missing fields, nulls, type mixes, malformed lines,
+empty lines, duplicate keys, unicode — is fair game.
+"""
+
+import json
+import random
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+KEYS = ["a", "b", "c", "req", "res", "x.y", "time", "lat", "msg"]
+STRINGS = ["GET", "PUT", "", "hello world", "héllo", "line\nbreak",
+           'quo"te', "back\\slash", "tab\there", "200", "1e3", "0x10",
+           "  12 ", "Infinity", "naan", "ünïcødé-αβγ", "a" * 120]
+
+
+def rand_value(rng, depth):
+    r = rng.random()
+    if depth < 3 and r < 0.15:
+        return {rng.choice(KEYS): rand_value(rng, depth + 1)
+                for _ in range(rng.randrange(3))}
+    if depth < 3 and r < 0.25:
+        return [rand_value(rng, depth + 1)
+                for _ in range(rng.randrange(3))]
+    if r < 0.40:
+        return rng.choice(STRINGS)
+    if r < 0.55:
+        return rng.randrange(-10000, 10000)
+    if r < 0.70:
+        return round(rng.uniform(-1e6, 1e6), 6)
+    if r < 0.78:
+        return rng.choice([1e-30, 2.5e3, 0.125, 1e15, -0.0])
+    if r < 0.86:
+        return rng.choice([True, False])
+    if r < 0.94:
+        return None
+    return rng.choice(["2014-05-01T12:34:56.789Z", "2014-05-01",
+                       "not a date", "2014-13-99"])
+
+
+def rand_record(rng):
+    rec = {}
+    for _ in range(rng.randrange(1, 7)):
+        rec[rng.choice(KEYS)] = rand_value(rng, 0)
+    return rec
+
+
+def rand_line(rng):
+    r = rng.random()
+    if r < 0.85:
+        return json.dumps(rec_or_scalar(rng),
+                          ensure_ascii=rng.random() < 0.5).encode()
+    if r < 0.90:
+        return b""  # empty line
+    if r < 0.95:
+        good = json.dumps(rand_record(rng)).encode()
+        return good[:rng.randrange(len(good))]  # truncated
+    return bytes(rng.randrange(33, 126)
+                 for _ in range(rng.randrange(1, 30)))  # garbage
+
+
+def rec_or_scalar(rng):
+    if rng.random() < 0.9:
+        return rand_record(rng)
+    return rand_value(rng, 0)
+
+
+def rand_query(rng):
+    from dragnet_amd.query import query_load
+    nbd = rng.randrange(0, 4)
+    parts = []
+    for _ in range(nbd):
+        k = rng.choice(KEYS + ["req.a", "res.b", "a.b"])
+        r = rng.random()
+        if r < 0.2:
+            parts.append(k + "[aggr=quantize]")
+        elif r < 0.35:
+            parts.append(k + "[aggr=lquantize,step=%d]"
+                         % rng.choice([10, 100, 1000]))
+        elif r < 0.45 and k in ("time",):
+            parts.append(k + "[date]")
+        else:
+            parts.append(k)
+    filt = None
+    if rng.random() < 0.6:
+        op = rng.choice(["eq", "ne", "lt", "le", "gt", "ge"])
+        k = rng.choice(KEYS + ["req.a"])
+        v = rng.choice(["GET", "200", 200, 0, True, None, "héllo",
+                        -5.5, "x"])
+        filt = {op: [k, v]}
+        if rng.random() < 0.3:
+            filt = {rng.choice(["and", "or"]):
+                    [filt, {"eq": [rng.choice(KEYS), rng.choice(
+                        ["PUT", 1, None])]}]}
+    return query_load(filter=filt,
+                      breakdown_specs=",".join(parts) or None)
+
+
+@pytest.mark.parametrize("seed", range(8))
+def test_fuzz_differential(seed, tmp_path):
+    import torch
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from dragnet_amd.engine.cpu import CpuEngine
+    from dragnet_amd.engine.gpu import GpuEngine
+
+    rng = random.Random(1000 + seed)
+    lines = [rand_line(rng) for _ in range(2000)]
+    path = tmp_path / ("fuzz%d.ndjson" % seed)
+    path.write_bytes(b"\n".join(lines) + b"\n")
+
+    cpu, gpu = CpuEngine(), GpuEngine()
+    for qi in range(6):
+        q = rand_query(rng)
+        c = cpu.scan([str(path)], [q])
+        g = gpu.scan([str(path)], [q])
+        desc = (seed, qi, q.filter, [b["name"] for b in q.breakdowns])
+        assert g.aggregators[0].points() == c.aggregators[0].points(), \
+            desc
+        cs = dict(c.stages)["json parser"]
+        gs = dict(g.stages)["json parser"]
+        assert gs == cs, desc

@@ -259,11 +259,22 @@ def compile_plan(queries, ds_filter=None, time_field=None,
                       if not (s in seen or seen.add(s))]
         synth_req.extend(my_synth_u)
 
+        # names materialized by this query's synthetic stage: any
+        # column with that NAME sees the converted unix seconds (the
+        # CPU pipeline mutates fields[name]; stream-synthetic.js:80)
+        synth_names = {b["name"]: b["field"]
+                       for b in q.breakdowns if "date" in b}
+        if q.time_field:
+            synth_names.setdefault(q.time_field, q.time_field)
+
         bd_off = len(bd_rows)
         for b in q.breakdowns:
             if "date" in b:
                 kind = 1
                 ref = synth_index(b["field"])
+            elif b["name"] in synth_names:
+                kind = 1
+                ref = synth_index(synth_names[b["name"]])
             else:
                 kind = 0
                 ref = fields.slot(b["name"])

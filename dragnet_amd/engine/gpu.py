@@ -239,7 +239,12 @@ class _ScanContext(object):
         into ping-pong pinned buffers (no intermediate Python bytes):
         each chunk is cut at its last newline and the tail is carried
         into the head of the other buffer while the GPU works on the
-        previous chunk."""
+        previous chunk.  Regular-file chunks are filled with parallel
+        preadv calls (page-cache reads are single-core bound at
+        ~15-25 GB/s; four readers roughly triple that)."""
+        import concurrent.futures as cf
+        import stat as _stat
+
         torch = self.t
         cap = self.eng.chunk_bytes
         if self._pinned is None:
@@ -249,6 +254,26 @@ class _ScanContext(object):
                             pin_memory=True)]
         views = [memoryview(p.numpy()) for p in pins]
         evs = [None, None]
+        pool = cf.ThreadPoolExecutor(max_workers=4)
+
+        def fill_from(fd, fpos, fsize, view, at, want, seq_file):
+            """Read up to `want` bytes of fd@fpos into view[at:].
+            Returns bytes read.  Parallel preadv for regular files."""
+            want = min(want, fsize - fpos)
+            if want <= 0:
+                return 0
+            if seq_file is not None:  # char device / pipe: sequential
+                return seq_file.readinto(view[at:at + want]) or 0
+            if want < (8 << 20):
+                return os.preadv(fd, [view[at:at + want]], fpos)
+            nsec = 4
+            sec = (want + nsec - 1) // nsec
+            futs = []
+            for s in range(0, want, sec):
+                e = min(s + sec, want)
+                futs.append(pool.submit(
+                    os.preadv, fd, [view[at + s:at + e]], fpos + s))
+            return sum(f.result() for f in futs)
 
         def last_newline(buf, n):
             probe = max(0, n - (1 << 16))
@@ -258,56 +283,76 @@ class _ScanContext(object):
             return bytes(buf[:probe]).rfind(b"\n")
 
         cur = 0
-        head = 0  # carried tail bytes at the front of pins[cur]
+        head = 0
         fiter = iter(files)
-        f = None
-        while True:
-            # pins[cur] may still be the source of an in-flight H2D
-            if evs[cur] is not None:
-                evs[cur].synchronize()
-                evs[cur] = None
-            n = head
-            eof = False
-            while n < cap:
-                if f is None:
-                    try:
-                        f = open(next(fiter), "rb", buffering=0)
-                    except StopIteration:
-                        eof = True
-                        break
-                got = f.readinto(views[cur][n:cap])
-                if not got:
-                    f.close()
-                    f = None
-                    continue
-                n += got
-            if n == 0:
-                break
-            cut = last_newline(views[cur], n)
-            if eof and cut < n - 1:
-                # final partial line: terminate it
-                views[cur][n:n + 1] = b"\n"
-                n += 1
-                cut = n - 1
-            if cut < 0:
-                if n >= cap:
-                    raise RuntimeError(
-                        "record exceeds the chunk buffer (%d bytes); "
-                        "raise DRAGNET_CHUNK_MB" % cap)
-                tail = n  # no newline yet: carry the whole buffer
-            else:
-                evs[cur] = self._run_pinned(pins[cur], cut + 1)
-                tail = n - (cut + 1)
-            nxt = 1 - cur
-            if tail:
-                if evs[nxt] is not None:
-                    evs[nxt].synchronize()
-                    evs[nxt] = None
-                views[nxt][:tail] = views[cur][cut + 1:n]
-            head = tail
-            cur = nxt
-            if eof:
-                break
+        fobj = None     # (fd, pos, size, seq_file or None)
+        try:
+            while True:
+                if evs[cur] is not None:
+                    evs[cur].synchronize()
+                    evs[cur] = None
+                n = head
+                eof = False
+                while n < cap:
+                    if fobj is None:
+                        try:
+                            path = next(fiter)
+                        except StopIteration:
+                            eof = True
+                            break
+                        st = os.stat(path)
+                        if _stat.S_ISREG(st.st_mode):
+                            fd = os.open(path, os.O_RDONLY)
+                            fobj = (fd, 0, st.st_size, None)
+                        else:
+                            sf = open(path, "rb", buffering=0)
+                            fobj = (sf.fileno(), 0, 1 << 62, sf)
+                    fd, fpos, fsize, sf = fobj
+                    got = fill_from(fd, fpos, fsize, views[cur], n,
+                                    cap - n, sf)
+                    if got <= 0:
+                        if sf is not None:
+                            sf.close()
+                        else:
+                            os.close(fd)
+                        fobj = None
+                        continue
+                    fobj = (fd, fpos + got, fsize, sf)
+                    n += got
+                if n == 0:
+                    break
+                cut = last_newline(views[cur], n)
+                if eof and cut < n - 1:
+                    views[cur][n:n + 1] = b"\n"
+                    n += 1
+                    cut = n - 1
+                if cut < 0:
+                    if n >= cap:
+                        raise RuntimeError(
+                            "record exceeds the chunk buffer (%d "
+                            "bytes); raise DRAGNET_CHUNK_MB" % cap)
+                    tail = n
+                else:
+                    evs[cur] = self._run_pinned(pins[cur], cut + 1)
+                    tail = n - (cut + 1)
+                nxt = 1 - cur
+                if tail:
+                    if evs[nxt] is not None:
+                        evs[nxt].synchronize()
+                        evs[nxt] = None
+                    views[nxt][:tail] = views[cur][cut + 1:n]
+                head = tail
+                cur = nxt
+                if eof:
+                    break
+        finally:
+            if fobj is not None:
+                _fd, _p, _s, sf = fobj
+                if sf is not None:
+                    sf.close()
+                else:
+                    os.close(_fd)
+            pool.shutdown(wait=False)
 
     def _run_pinned(self, pin, n):
         """H2D + kernels for pin[:n] (already newline-terminated)."""

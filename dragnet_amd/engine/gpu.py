@@ -254,7 +254,8 @@ class _ScanContext(object):
                             pin_memory=True)]
         views = [memoryview(p.numpy()) for p in pins]
         evs = [None, None]
-        pool = cf.ThreadPoolExecutor(max_workers=4)
+        pool = cf.ThreadPoolExecutor(
+            max_workers=_env_int("DRAGNET_READERS", 8))
 
         def fill_from(fd, fpos, fsize, view, at, want, seq_file):
             """Read up to `want` bytes of fd@fpos into view[at:].
@@ -266,7 +267,7 @@ class _ScanContext(object):
                 return seq_file.readinto(view[at:at + want]) or 0
             if want < (8 << 20):
                 return os.preadv(fd, [view[at:at + want]], fpos)
-            nsec = 4
+            nsec = _env_int("DRAGNET_READERS", 8)
             sec = (want + nsec - 1) // nsec
             futs = []
             for s in range(0, want, sec):

@@ -58,7 +58,7 @@ class GpuEngine(object):
         self.ops = load_ops(required=True)
         self.device = device if device is not None else \
             torch.device("cuda", torch.cuda.current_device())
-        self.chunk_bytes = _env_int("DRAGNET_CHUNK_MB", 64) * 1024 * 1024
+        self.chunk_bytes = _env_int("DRAGNET_CHUNK_MB", 256) * 1024 * 1024
 
     # ---- public engine interface ----
 
@@ -255,7 +255,7 @@ class _ScanContext(object):
         views = [memoryview(p.numpy()) for p in pins]
         evs = [None, None]
         pool = cf.ThreadPoolExecutor(
-            max_workers=_env_int("DRAGNET_READERS", 8))
+            max_workers=_env_int("DRAGNET_READERS", 4))
 
         def fill_from(fd, fpos, fsize, view, at, want, seq_file):
             """Read up to `want` bytes of fd@fpos into view[at:].
@@ -267,7 +267,7 @@ class _ScanContext(object):
                 return seq_file.readinto(view[at:at + want]) or 0
             if want < (8 << 20):
                 return os.preadv(fd, [view[at:at + want]], fpos)
-            nsec = _env_int("DRAGNET_READERS", 8)
+            nsec = _env_int("DRAGNET_READERS", 4)
             sec = (want + nsec - 1) // nsec
             futs = []
             for s in range(0, want, sec):

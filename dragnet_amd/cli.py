@@ -269,9 +269,6 @@ def _get_datasource(cfg, name):
     return ds
 
 
-def _make_engine(opts):
-    from .engine import get_engine
-    return get_engine()
 
 
 # ---- datasource commands ----

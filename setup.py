@@ -34,7 +34,7 @@ setup(
             sources=["dragnet_amd/ops/hip/ext.hip"],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
-                "nvcc": ["-O3", "-std=c++17"]
+                "nvcc": [os.environ.get("DN_OPT", "-O3"), "-std=c++17"]
                         + (["-DDN_DIRECT_BYTES"]
                            if os.environ.get("DN_DIRECT_BYTES") else []),
             },

@@ -5,7 +5,6 @@ the DP/RCCL path made correct by construction (associative point
 merge), exercised here without a GPU.
 """
 
-import json
 import os
 import sys
 

@@ -71,14 +71,21 @@ def main():
 
     import torch
     assert torch.cuda.is_available(), "bench needs a GPU"
-    torch.cuda.set_device(local_rank)
+
+    # backend override for plumbing rehearsals on a single GPU
+    # (DRAGNET_BENCH_BACKEND=gloo shares cuda:0 across ranks and
+    # merges through CPU tensors; production N>1 uses RCCL)
+    backend = os.environ.get("DRAGNET_BENCH_BACKEND", "nccl")
+    ngpu = torch.cuda.device_count()
+    dev_index = local_rank if backend == "nccl" else local_rank % ngpu
+    torch.cuda.set_device(dev_index)
 
     dist = None
     if world > 1:
         import torch.distributed as torch_dist
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29601")
-        torch_dist.init_process_group("nccl", rank=rank,
+        torch_dist.init_process_group(backend, rank=rank,
                                       world_size=world)
         dist = torch_dist
 
@@ -98,7 +105,7 @@ def main():
     log("pool: %d records, %.1f MB in %.1fs" % (
         nrec, pool_bytes / 1e6, time.time() - t0))
 
-    eng = GpuEngine(device=torch.device("cuda", local_rank))
+    eng = GpuEngine(device=torch.device("cuda", dev_index))
     cplan = planmod.compile_plan([query])
     eng.chunk_bytes = pool_bytes  # single-chunk pool
     ctx = _ScanContext(eng, cplan,
@@ -128,7 +135,9 @@ def main():
             ctx.scan_resident(h2d=not args.device_resident)
         aggs, _stages = ctx.finalize([query])
         if dist is not None:
-            merged = merge_tables_tensor(aggs[0], query, device)
+            merged = merge_tables_tensor(
+                aggs[0], query,
+                device if backend == "nccl" else torch.device("cpu"))
         else:
             merged = aggs[0]
         return merged

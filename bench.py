@@ -161,7 +161,8 @@ def main():
 
     # max over ranks
     if dist is not None:
-        t = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        red_dev = device if backend == "nccl" else torch.device("cpu")
+        t = torch.tensor([elapsed], dtype=torch.float64, device=red_dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 

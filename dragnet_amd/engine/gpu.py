@@ -257,24 +257,45 @@ class _ScanContext(object):
         pool = cf.ThreadPoolExecutor(
             max_workers=_env_int("DRAGNET_READERS", 4))
 
+        def pread_full(fd, mv, off):
+            """preadv until mv is full (a single preadv may legally
+            return short; a short section would leave a HOLE of stale
+            bytes that corrupts record framing — observed at the
+            multi-GB scale)."""
+            done = 0
+            while done < len(mv):
+                got = os.preadv(fd, [mv[done:]], off + done)
+                if got <= 0:
+                    break
+                done += got
+            return done
+
         def fill_from(fd, fpos, fsize, view, at, want, seq_file):
             """Read up to `want` bytes of fd@fpos into view[at:].
-            Returns bytes read.  Parallel preadv for regular files."""
+            Returns bytes read (contiguous).  Parallel preadv for
+            regular files."""
             want = min(want, fsize - fpos)
             if want <= 0:
                 return 0
             if seq_file is not None:  # char device / pipe: sequential
                 return seq_file.readinto(view[at:at + want]) or 0
             if want < (8 << 20):
-                return os.preadv(fd, [view[at:at + want]], fpos)
+                return pread_full(fd, view[at:at + want], fpos)
             nsec = _env_int("DRAGNET_READERS", 4)
             sec = (want + nsec - 1) // nsec
             futs = []
             for s in range(0, want, sec):
                 e = min(s + sec, want)
-                futs.append(pool.submit(
-                    os.preadv, fd, [view[at + s:at + e]], fpos + s))
-            return sum(f.result() for f in futs)
+                futs.append((e - s, pool.submit(
+                    pread_full, fd, view[at + s:at + e], fpos + s)))
+            # only the contiguous prefix counts
+            total = 0
+            for length, fut in futs:
+                got = fut.result()
+                total += got
+                if got < length:
+                    break
+            return total
 
         def last_newline(buf, n):
             probe = max(0, n - (1 << 16))

@@ -348,6 +348,7 @@ class _ScanContext(object):
                     views[cur][n:n + 1] = b"\n"
                     n += 1
                     cut = n - 1
+                nxt = 1 - cur
                 if cut < 0:
                     if n >= cap:
                         raise RuntimeError(
@@ -355,14 +356,18 @@ class _ScanContext(object):
                             "bytes); raise DRAGNET_CHUNK_MB" % cap)
                     tail = n
                 else:
-                    evs[cur] = self._run_pinned(pins[cur], cut + 1)
                     tail = n - (cut + 1)
-                nxt = 1 - cur
+                # the tail must be carried over BEFORE the chunk is
+                # launched: _run_pinned pads pin[cut+1:] with newlines
+                # and would clobber the carried bytes (record-splitting
+                # corruption observed at multi-GB scale)
                 if tail:
                     if evs[nxt] is not None:
                         evs[nxt].synchronize()
                         evs[nxt] = None
-                    views[nxt][:tail] = views[cur][cut + 1:n]
+                    views[nxt][:tail] = views[cur][n - tail:n]
+                if cut >= 0:
+                    evs[cur] = self._run_pinned(pins[cur], cut + 1)
                 head = tail
                 cur = nxt
                 if eof:

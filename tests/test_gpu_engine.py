@@ -279,3 +279,28 @@ def test_deterministic_across_runs(engines, tmp_path):
     a = gpu.scan([str(path)], [q]).aggregators[0].points()
     b = gpu.scan([str(path)], [q]).aggregators[0].points()
     assert a == b
+
+
+def test_many_chunk_boundaries(engines, tmp_path):
+    """Tiny chunks force hundreds of boundary tail-carries; record
+    framing must survive every one (regression: the chunk padding once
+    clobbered carried tail bytes at multi-GB scale)."""
+    cpu, gpu = engines
+    from dragnet_amd.query import query_load
+    from dragnet_amd.tools.mktestdata import generate_lines
+    path = tmp_path / "bounds.ndjson"
+    with open(path, "wb") as f:
+        for line in generate_lines(20_000, seed=17):
+            f.write(line)
+    q = query_load(breakdown_specs="req.method,res.statusCode")
+    old = gpu.chunk_bytes
+    try:
+        gpu.chunk_bytes = 1 << 14  # 16 KB chunks -> ~280 boundaries
+        g = gpu.scan([str(path)], [q])
+    finally:
+        gpu.chunk_bytes = old
+    c = cpu.scan([str(path)], [q])
+    gs = dict(g.stages)["json parser"]
+    assert gs["ninputs"] == 20_000
+    assert gs["invalid json"] == 0
+    assert g.aggregators[0].points() == c.aggregators[0].points()

@@ -58,9 +58,35 @@ def fnv1a(data, h=FNV_OFFSET):
     return h
 
 
+def _mix64(x):
+    x &= MASK64
+    x ^= x >> 33
+    x = (x * 0xFF51AFD7ED558CCD) & MASK64
+    x ^= x >> 33
+    x = (x * 0xC4CEB9FE1A85EC53) & MASK64
+    x ^= x >> 33
+    return x
+
+
+def comp_hash(b):
+    """Hash of one path component over zero-padded 8-byte words —
+    computable on-device with SWAR window loads (scan_kernels.hip
+    scan_key_sig must produce identical values)."""
+    h = 0x9E3779B97F4A7C15
+    for k in range(0, max(len(b), 1), 8):
+        w = int.from_bytes(b[k:k + 8].ljust(8, b"\0"), "little")
+        h = _mix64(h ^ w)
+    return _mix64(h ^ len(b))
+
+
 def path_sig(path):
-    """Signature of a dotted path: FNV-1a over the path bytes."""
-    return fnv1a(path.encode("utf-8"))
+    """Signature of a dotted path: a chained component hash.  Dots
+    split components, so a literal "a.b" key and nested a->b produce
+    the same signature (the aggregation lookup rule)."""
+    sig = FNV_OFFSET
+    for comp in path.split("."):
+        sig = _mix64(sig ^ comp_hash(comp.encode("utf-8")))
+    return sig
 
 
 class PlanError(Exception):

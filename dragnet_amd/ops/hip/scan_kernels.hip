@@ -216,6 +216,70 @@ DEV bool scan_string_fast(Cursor& c, uint32_t& off_out,
   }
 }
 
+// Component hash over zero-padded 8-byte words (mirrors
+// plan.comp_hash; any overread is masked away and stays inside the
+// padded buffer).
+DEV uint64_t comp_hash_span(Bytes B, uint32_t s, uint32_t e) {
+  uint64_t h = 0x9E3779B97F4A7C15ull;
+  uint32_t len = e - s;
+  uint32_t k = 0;
+  do {
+    uint64_t w;
+    __builtin_memcpy(&w, B.ptr(s + k), 8);
+    uint32_t rem = len - k;
+    if (rem < 8)
+      w = (rem == 0) ? 0ull : (w & ((~0ull) >> (8 * (8 - rem))));
+    h = mix64(h ^ w);
+    k += 8;
+  } while (k < len);
+  return mix64(h ^ (uint64_t)len);
+}
+
+// Scan a KEY (cursor after the opening quote): SWAR windows to the
+// next quote/dot/escape/control; chains component hashes into the
+// path signature (mirrors plan.path_sig — '.' splits components so
+// literal dotted keys unify with nested paths).  Escaped keys are
+// validated but get a sentinel signature that matches no field.
+DEV int scan_key_sig(Cursor& c, uint64_t parent, uint64_t& sig_out) {
+  const uint8_t* d = c.B.mem - c.B.bias;
+  uint32_t p = c.pos, end = c.end;
+  uint32_t comp_s = p;
+  uint64_t sig = parent;
+  while (true) {
+    while (p + 8 <= end) {
+      uint64_t w;
+      __builtin_memcpy(&w, d + p, 8);
+      uint64_t m = str_special_mask(w) |
+                   hz8(w ^ 0x2E2E2E2E2E2E2E2Eull);
+      if (m == 0) { p += 8; continue; }
+      p += ((uint32_t)__ffsll((unsigned long long)m) - 1) >> 3;
+      break;
+    }
+    if (p >= end) return 0;
+    uint8_t b = d[p];
+    if (b == '"') {
+      sig_out = mix64(sig ^ comp_hash_span(c.B, comp_s, p));
+      c.pos = p + 1;
+      return 1;
+    }
+    if (b == '.') {
+      sig = mix64(sig ^ comp_hash_span(c.B, comp_s, p));
+      p++;
+      comp_s = p;
+      continue;
+    }
+    if (b == '\\') {
+      c.pos = p;
+      uint32_t o, l;
+      if (!scan_string_fast(c, o, l)) return 0;
+      sig_out = 0x1ull;  // never matches a compiled signature
+      return 1;
+    }
+    if (b < 0x20) return 0;
+    p++;  // near-record-end tail: plain byte
+  }
+}
+
 // -------------------------------------------------------------------
 // per-record extracted field values, stored in LDS (SoA, [field][tid])
 
@@ -516,48 +580,11 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
 
   // scan a JSON string starting AFTER the opening quote; returns false
   // on bad escape/unterminated; sets len (raw bytes), computes fnv
-  auto scan_string = [&](uint64_t fnv_in, uint64_t& fnv_out,
-                         uint32_t& off_out, uint32_t& len_out) -> bool {
-    uint32_t off = c.pos;
-    uint64_t h = fnv_in;
-    while (true) {
-      if (c.eof()) return false;
-      uint8_t b = c.next();
-      if (b == '"') { off_out = off; len_out = c.pos - 1 - off; fnv_out = h; return true; }
-      if (b == '\\') {
-        h = fnv1a_byte(h, b);
-        if (c.eof()) return false;
-        uint8_t e = c.next();
-        h = fnv1a_byte(h, e);
-        if (e == 'u') {
-          for (int k = 0; k < 4; k++) {
-            if (c.eof()) return false;
-            uint8_t x = c.next();
-            bool hex = (x >= '0' && x <= '9') || (x >= 'a' && x <= 'f') || (x >= 'A' && x <= 'F');
-            if (!hex) return false;
-            h = fnv1a_byte(h, x);
-          }
-        } else if (!(e=='"'||e=='\\'||e=='/'||e=='b'||e=='f'||e=='n'||e=='r'||e=='t')) {
-          return false;
-        }
-      } else if (b < 0x20) {
-        return false;  // raw control char in string
-      } else {
-        h = fnv1a_byte(h, b);
-      }
-    }
-  };
-
   // parse the key of an object member (cursor at '"'), extending the
-  // parent signature: sig = fnv(parent [+ '.'] + keybytes)
+  // parent signature via chained component hashes (plan.path_sig)
   auto parse_key = [&](uint64_t parent, bool root, uint64_t& sig_out) -> bool {
     if (c.eof() || c.next() != '"') return false;
-    uint64_t base = root ? FNV_OFFSET : fnv1a_byte(parent, (uint8_t)'.');
-    uint32_t off, len;
-    uint64_t h;
-    if (!scan_string(base, h, off, len)) return false;
-    sig_out = h;
-    return true;
+    return scan_key_sig(c, root ? FNV_OFFSET : parent, sig_out) != 0;
   };
 
   // Main loop: parse values iteratively.

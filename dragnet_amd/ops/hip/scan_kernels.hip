@@ -316,142 +316,64 @@ DEV double scale10(double v, long ex) {
 
 struct NumOut { double v; bool ok; };
 
-// ---- SWAR decimal parsing (simdjson-style 8-digit blocks) ----
-
-// 8 ASCII digits in memory order (low byte = first/most-significant)
-DEV uint64_t parse8digits(uint64_t w) {
-  w -= 0x3030303030303030ull;
-  w = (w * 10) + (w >> 8);
-  w = (((w & 0x000000FF000000FFull) * ((100ull << 32) + 1)) >> 32)
-          * 10000
-      + ((((w >> 16) & 0x000000FF000000FFull)
-          * ((100ull << 32) + 1)) >> 32);
-  return w;
-}
-
-// EXACT per-byte mask: high bit set where the byte is an ASCII digit
-// (nibble arithmetic stays within bytes — no inter-byte carries)
-DEV uint64_t digit_mask8(uint64_t w) {
-  uint64_t lo = w & 0x0F0F0F0F0F0F0F0Full;
-  uint64_t hi = (w >> 4) & 0x0F0F0F0F0F0F0F0Full;
-  uint64_t h3 = hi ^ 0x0303030303030303ull;
-  uint64_t nz3 = ((h3 + 0x7F7F7F7F7F7F7F7Full)
-                  & 0x8080808080808080ull);           // high nibble != 3
-  uint64_t gt9 = ((lo + 0x0606060606060606ull)
-                  & 0x1010101010101010ull) << 3;      // low nibble > 9
-  return 0x8080808080808080ull & ~nz3 & ~gt9;
-}
-
-__device__ __constant__ unsigned long long P10I[8] = {
-  1ull, 10ull, 100ull, 1000ull, 10000ull, 100000ull, 1000000ull,
-  10000000ull};
-
-// length of the digit run starting at s (capped at end)
-DEV uint32_t digit_run(const uint8_t* d, uint32_t s, uint32_t end) {
-  uint32_t p = s;
-  while (p + 8 <= end) {
-    uint64_t w;
-    __builtin_memcpy(&w, d + p, 8);
-    uint64_t nd = ~digit_mask8(w) & 0x8080808080808080ull;
-    if (nd)
-      return p + (((uint32_t)__ffsll((unsigned long long)nd) - 1) >> 3)
-             - s;
-    p += 8;
-  }
-  while (p < end && d[p] >= '0' && d[p] <= '9') p++;
-  return p - s;
-}
-
-// value of digits d[s..s+len), len <= 19
-DEV uint64_t parse_digits(const uint8_t* d, uint32_t s, uint32_t len) {
-  uint64_t v = 0;
-  uint32_t k = 0;
-  for (; k + 8 <= len; k += 8) {
-    uint64_t w;
-    __builtin_memcpy(&w, d + s + k, 8);
-    v = v * 100000000ull + parse8digits(w);
-  }
-  uint32_t t = len - k;
-  if (t) {
-    uint64_t tv;
-    if (s + len >= 8) {
-      // end-aligned load: tail digits sit in the HIGH lanes; replace
-      // the low (earlier-garbage) lanes with '0' = leading zeros
-      uint64_t w;
-      __builtin_memcpy(&w, d + s + len - 8, 8);
-      uint64_t keep = (~0ull) << ((8 - t) * 8);
-      w = (w & keep) | (0x3030303030303030ull & ~keep);
-      tv = parse8digits(w);
-    } else {
-      tv = 0;
-      for (uint32_t j = 0; j < t; j++)
-        tv = tv * 10 + (d[s + k + j] - '0');
-    }
-    v = v * P10I[t] + tv;
-  }
-  return v;
-}
-
 DEV NumOut parse_json_number(Cursor& c) {
   NumOut out; out.ok = false; out.v = 0.0;
-  const uint8_t* d = c.B.mem - c.B.bias;
-  uint32_t p = c.pos, end = c.end;
   bool neg = false;
-  if (p < end && d[p] == '-') { neg = true; p++; }
-  if (p >= end) return out;
-
+  if (!c.eof() && c.peek() == '-') { neg = true; c.pos++; }
+  if (c.eof()) return out;
   // integer part: 0 | [1-9][0-9]*
-  uint32_t L0 = digit_run(d, p, end);
-  if (L0 == 0) return out;
-  if (d[p] == '0' && L0 > 1) return out;  // leading-zero rule
-  uint32_t s0 = p;
-  p += L0;
-
-  uint32_t s1 = 0, L1 = 0;
-  if (p < end && d[p] == '.') {
-    p++;
-    L1 = digit_run(d, p, end);
-    if (L1 == 0) return out;
-    s1 = p;
-    p += L1;
-  }
-
-  long e10 = 0;
-  int esign = 1;
-  if (p < end && (d[p] == 'e' || d[p] == 'E')) {
-    p++;
-    if (p < end && (d[p] == '+' || d[p] == '-')) {
-      if (d[p] == '-') esign = -1;
-      p++;
+  uint64_t mant = 0;
+  int ndig = 0, extra_exp = 0;
+  uint8_t b = c.peek();
+  if (b == '0') {
+    c.pos++; ndig = 1;
+    if (!c.eof()) { uint8_t nb = c.peek(); if (nb >= '0' && nb <= '9') return out; }
+  } else if (b >= '1' && b <= '9') {
+    while (!c.eof()) {
+      uint8_t d = c.peek();
+      if (d < '0' || d > '9') break;
+      c.pos++;
+      if (ndig < 19) { mant = mant * 10u + (d - '0'); ndig++; }
+      else extra_exp++;
     }
-    uint32_t Le = digit_run(d, p, end);
-    if (Le == 0) return out;
-    for (uint32_t k = 0; k < Le; k++)
-      if (e10 < 100000) e10 = e10 * 10 + (d[p + k] - '0');
-    p += Le;
+  } else {
+    return out;
   }
-
-  // mantissa: the first 19 significant digits of int||frac; dropped
-  // int digits shift the exponent up, consumed frac digits down
-  uint32_t take0 = L0 < 19u ? L0 : 19u;
-  uint64_t mant = parse_digits(d, s0, take0);
-  long extra = (long)L0 - (long)take0;
-  uint32_t rem = 19u - take0;
-  uint32_t take1 = L1 < rem ? L1 : rem;
-  if (take1) {
-    uint64_t fv = parse_digits(d, s1, take1);
-    uint64_t scale = 1;
-    uint32_t q = take1;
-    while (q >= 8) { scale *= 100000000ull; q -= 8; }
-    scale *= P10I[q];
-    mant = mant * scale + fv;
+  // fraction
+  if (!c.eof() && c.peek() == '.') {
+    c.pos++;
+    int fdig = 0;
+    while (!c.eof()) {
+      uint8_t d = c.peek();
+      if (d < '0' || d > '9') break;
+      c.pos++;
+      if (ndig < 19) { mant = mant * 10u + (d - '0'); ndig++; extra_exp--; }
+      fdig++;
+    }
+    if (fdig == 0) return out;
   }
-  extra -= (long)take1;
-
-  double v = scale10((double)mant, esign * e10 + extra);
+  // exponent
+  int esign = 1; long e10 = 0;
+  if (!c.eof() && (c.peek() == 'e' || c.peek() == 'E')) {
+    c.pos++;
+    if (!c.eof() && (c.peek() == '+' || c.peek() == '-')) {
+      if (c.peek() == '-') esign = -1;
+      c.pos++;
+    }
+    int edig = 0;
+    while (!c.eof()) {
+      uint8_t d = c.peek();
+      if (d < '0' || d > '9') break;
+      c.pos++;
+      if (e10 < 100000) e10 = e10 * 10 + (d - '0');
+      edig++;
+    }
+    if (edig == 0) return out;
+  }
+  long exp10 = esign * e10 + extra_exp;
+  double v = scale10((double)mant, exp10);
   out.v = neg ? -v : v;
   out.ok = true;
-  c.pos = p;
   return out;
 }
 

@@ -154,7 +154,6 @@ _OPTDEFS = {
     "warnings": ("--warnings", False),
     "dry-run": ("--dry-run", False),
     "verbose": ("-v", False),
-    "gpus": ("--gpus", True),
 }
 
 _SHORT = {"-f": "filter", "-b": "breakdowns", "-v": "verbose",
@@ -486,7 +485,7 @@ def _output_result(query, opts, result, title=None):
 def cmd_scan(argv):
     opts, args = parse_args(argv, [
         "before", "after", "filter", "breakdowns", "raw", "points",
-        "counters", "warnings", "gnuplot", "dry-run", "gpus"])
+        "counters", "warnings", "gnuplot", "dry-run"])
     check_arg_count(args, 1)
     cfg = _load_config()
     ds = _get_datasource(cfg, args[0])
@@ -556,7 +555,7 @@ def _metrics_for_index(cfg, ds, opts):
 
 def cmd_build(argv):
     opts, args = parse_args(argv, [
-        "before", "after", "interval", "index-config", "dry-run", "gpus"])
+        "before", "after", "interval", "index-config", "dry-run"])
     check_arg_count(args, 1)
     cfg = _load_config()
     ds = _get_datasource(cfg, args[0])

@@ -127,24 +127,46 @@ def main():
         log("hipGraph capture: %s"
             % ("ok" if graph is not None else "unavailable"))
 
+    # Software pipeline: each step enqueues its scan + extraction and
+    # then decodes the PREVIOUS step's snapshot on the host while this
+    # step's copies/kernels run (decode syncs only on the snapshot's
+    # event, on a dedicated D2H stream).  drain() decodes the last
+    # pending handle, so a timed region of K steps + one drain does
+    # exactly K scans and K decodes — no work leaves the clock.
+    pending = [None]
+
+    def do_merge(agg):
+        if dist is not None:
+            return merge_tables_tensor(
+                agg, query,
+                device if backend == "nccl" else torch.device("cpu"))
+        return agg
+
     def step():
         if graph is not None:
             graph.replay()
         else:
             ctx.reset()
             ctx.scan_resident(h2d=not args.device_resident)
-        aggs, _stages = ctx.finalize([query])
-        if dist is not None:
-            merged = merge_tables_tensor(
-                aggs[0], query,
-                device if backend == "nccl" else torch.device("cpu"))
-        else:
-            merged = aggs[0]
+        ex = ctx.extract_async([query])
+        merged = None
+        if pending[0] is not None:
+            aggs, _stages = ctx.decode_extracted(pending[0], [query])
+            merged = do_merge(aggs[0])
+        pending[0] = ex
         return merged
+
+    def drain():
+        if pending[0] is None:
+            return None
+        aggs, _stages = ctx.decode_extracted(pending[0], [query])
+        pending[0] = None
+        return do_merge(aggs[0])
 
     # warmup
     for _ in range(args.warmup):
-        result = step()
+        step()
+    result = drain()
     assert int(ctx.counters[1].item()) == 0, "invalid JSON in pool?!"
 
     # timed region
@@ -153,7 +175,10 @@ def main():
     torch.cuda.synchronize(device)
     t_start = time.time()
     for _ in range(args.steps):
-        result = step()
+        r = step()
+        if r is not None:
+            result = r
+    result = drain()
     torch.cuda.synchronize(device)
     if dist is not None:
         dist.barrier()

@@ -170,6 +170,8 @@ class _ScanContext(object):
         self._pinned = None
         self._copy_ev = None  # guards pinned-buffer reuse across chunks
         self.agg_slots = agg_slots
+        self.dict_slots = dict_slots
+        self._fin_stream = None  # lazy: D2H lane for pipelined decode
 
     # ---- chunk feeding ----
 
@@ -536,39 +538,79 @@ class _ScanContext(object):
     # ---- results ----
 
     def finalize(self, queries):
-        torch = self.t
-        torch.cuda.synchronize(self.eng.device)
-        cnt = self.counters.cpu().numpy()
+        return self.decode_extracted(self.extract_async(queries),
+                                     queries)
 
-        # dictionaries
-        n_str = int(self.sd["next"].item())
-        n_num = int(self.nd["next"].item())
-        strings = []
-        if n_str:
-            off, ln = self.eng.ops.extract_strdict(
-                self.sd["state"], self.sd["hash"], self.sd["id"],
-                self.sd["off"], self.sd["len"], self.sd["data"],
-                self.sd["used"], self.sd["next"], n_str)
-            off = off.cpu().numpy().astype(np.uint32)
-            ln = ln.cpu().numpy().astype(np.uint32)
-            used = int(self.sd["used"].item())
-            blob = self.sd["data"][:used].cpu().numpy().tobytes()
-            for i in range(n_str):
-                raw = blob[off[i]:off[i] + ln[i]]
-                strings.append(_decode_json_string(raw))
-        numbers = np.zeros(0)
-        if n_num:
-            numbers = self.eng.ops.extract_numdict(
-                self.nd["state"], self.nd["bits"], self.nd["id"],
-                self.nd["next"], n_num).cpu().numpy()
+    def extract_async(self, queries):
+        """Enqueue the table/dictionary extraction kernels and snapshot
+        everything decode needs into fresh device tensors, WITHOUT any
+        host sync.  The returned handle is safe across a subsequent
+        reset()/scan of this context (the snapshots are ordered on the
+        current stream before the reset's zeroing), which lets a caller
+        software-pipeline host-side decode of step k with step k+1's
+        copies/kernels (bench.py streaming mode)."""
+        h = {"cnt": self.counters.clone(),
+             "n_str": self.sd["next"].clone(),
+             "n_num": self.nd["next"].clone(),
+             "used": self.sd["used"].clone(),
+             # the dictionary byte blob is bump-allocated from 0 each
+             # scan, so the NEXT step overwrites it — snapshot it now
+             "blob": self.sd["data"].clone()}
+        h["str_off"], h["str_len"] = self.eng.ops.extract_strdict(
+            self.sd["state"], self.sd["hash"], self.sd["id"],
+            self.sd["off"], self.sd["len"], self.sd["data"],
+            self.sd["used"], self.sd["next"], self.dict_slots)
+        h["numbers"] = self.eng.ops.extract_numdict(
+            self.nd["state"], self.nd["bits"], self.nd["id"],
+            self.nd["next"], self.dict_slots)
+        h["aggs"] = []
+        for m, _q in enumerate(queries):
+            state, keys, count = self.tables[m]
+            h["aggs"].append(self.eng.ops.extract_agg_async(
+                state, keys, count, self.agg_slots))
+        torch = self.t
+        h["ev"] = torch.cuda.Event()
+        h["ev"].record()
+        return h
+
+    def decode_extracted(self, h, queries):
+        """Host-side half of finalize(): wait for the extraction
+        snapshot (ONLY — via its event, on a dedicated D2H stream, so
+        in-flight work from a subsequent step is not drained) and
+        decode into Aggregators."""
+        torch = self.t
+        if self._fin_stream is None:
+            self._fin_stream = torch.cuda.Stream(device=self.eng.device)
+        with torch.cuda.stream(self._fin_stream):
+            self._fin_stream.wait_event(h["ev"])
+            cnt = h["cnt"].cpu().numpy()
+
+            # dictionaries
+            n_str = int(h["n_str"].item())
+            n_num = int(h["n_num"].item())
+            strings = []
+            if n_str:
+                off = h["str_off"][:n_str].cpu().numpy().astype(np.uint32)
+                ln = h["str_len"][:n_str].cpu().numpy().astype(np.uint32)
+                used = int(h["used"].item())
+                blob = h["blob"][:used].cpu().numpy().tobytes()
+                for i in range(n_str):
+                    raw = blob[off[i]:off[i] + ln[i]]
+                    strings.append(_decode_json_string(raw))
+            numbers = np.zeros(0)
+            if n_num:
+                numbers = h["numbers"][:n_num].cpu().numpy()
+            hostk = []
+            for m, _q in enumerate(queries):
+                k_full, c_full, out_n = h["aggs"][m]
+                n_out = min(int(out_n.item()), self.agg_slots)
+                hostk.append((k_full[:n_out].cpu().numpy(),
+                              c_full[:n_out].cpu().numpy()))
 
         aggs = []
         for m, q in enumerate(queries):
-            state, keys, count = self.tables[m]
-            k, c = self.eng.ops.extract_agg(
-                state, keys, count, self.agg_slots)
-            k = k.cpu().numpy().astype(np.uint32)
-            c = c.cpu().numpy()
+            k, c = hostk[m]
+            k = k.astype(np.uint32)
             agg = Aggregator(q)
             mc = cnt[C_GLOBAL_N + m * CM_N:
                      C_GLOBAL_N + (m + 1) * CM_N]

@@ -243,6 +243,30 @@ std::vector<torch::Tensor> extract_agg(torch::Tensor state,
   return {out_keys.narrow(0, 0, n), out_counts.narrow(0, 0, n)};
 }
 
+// Async variant: no host sync — returns the full-capacity buffers plus
+// the device count; the caller slices after its own (later) sync.
+std::vector<torch::Tensor> extract_agg_async(torch::Tensor state,
+                                             torch::Tensor keys,
+                                             torch::Tensor count,
+                                             int64_t max_out) {
+  CHECK_GPU(state);
+  AggTable T = make_table(state, keys, count);
+  auto i32 = torch::dtype(torch::kInt32).device(state.device());
+  auto f64 = torch::dtype(torch::kFloat64).device(state.device());
+  auto out_keys = torch::zeros({max_out, (long)MAX_KEY}, i32);
+  auto out_counts = torch::zeros({max_out}, f64);
+  auto out_n = torch::zeros({1}, i32);
+  uint32_t blocks = (T.nslots + 255) / 256;
+  hipLaunchKernelGGL(extract_agg_kernel, dim3(blocks), dim3(256), 0,
+                     current_stream(), T, (uint32_t*)out_keys.data_ptr(),
+                     (double*)out_counts.data_ptr(),
+                     (uint32_t*)out_n.data_ptr(), (uint32_t)max_out);
+  hipError_t err = hipGetLastError();
+  TORCH_CHECK(err == hipSuccess, "extract_agg_async launch failed: ",
+              hipGetErrorString(err));
+  return {out_keys, out_counts, out_n};
+}
+
 std::vector<torch::Tensor> extract_strdict(
     torch::Tensor sd_state, torch::Tensor sd_hash, torch::Tensor sd_id,
     torch::Tensor sd_off, torch::Tensor sd_len, torch::Tensor sd_data,
@@ -291,6 +315,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("newline_index", &dn::newline_index, "device-side newline index");
   m.def("agg_descs_host", &dn::agg_descs_host);
   m.def("extract_agg", &dn::extract_agg);
+  m.def("extract_agg_async", &dn::extract_agg_async);
   m.def("extract_strdict", &dn::extract_strdict);
   m.def("extract_numdict", &dn::extract_numdict);
   m.attr("MAX_KEY") = (int)dn::MAX_KEY;

@@ -9,7 +9,9 @@ namespace dn {
 
 // ---- limits ----
 constexpr int MAX_FIELDS = 24;   // distinct dotted paths per plan
-constexpr int MAX_DEPTH  = 12;   // JSON nesting depth
+constexpr int MAX_DEPTH  = 64;   // validated JSON nesting depth
+constexpr int SLOT_DEPTH = 12;   // capture-slot tracking depth (deeper
+                                 // containers are inside arrays: never captured)
 constexpr int MAX_KEY    = 8;    // breakdown columns per metric
 constexpr int MAX_SYNTH  = 8;    // synthetic date fields per plan
 constexpr int PRED_STACK = 8;    // and/or nesting depth

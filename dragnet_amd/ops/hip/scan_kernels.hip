@@ -561,7 +561,7 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
   #define sig_stack_at(d) sig_lds[(d) * BLOCK + (uint32_t)threadIdx.x]
 
   uint64_t cont_slot = ~0ull;  // 5 bits/depth: captured slot or 31
-  uint32_t is_arr_bits = 0;       // bit d: container at depth d is array
+  uint64_t is_arr_bits = 0;       // bit d: container at depth d is array
   int depth = 0;                  // container depth (0 = at top value)
   int arr_depth = 0;              // number of array containers on stack
   uint64_t cur_sig = 0;           // path sig for the value being parsed
@@ -618,14 +618,14 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
           // fall through to "after value"
         } else {
           // push object frame
-          {
+          if (depth < SLOT_DEPTH) {
             uint64_t cs = (cur_capture && arr_depth == 0)
                               ? (uint64_t)cur_slot : 31ull;
             cont_slot = (cont_slot & ~(31ull << (5 * depth)))
                         | (cs << (5 * depth));
           }
           if (depth < SIG_DEPTH) sig_stack_at(depth) = cur_sig;
-          is_arr_bits &= ~(1u << depth);
+          is_arr_bits &= ~(1ull << depth);
           depth++;
           // parse first key
           uint64_t ksig;
@@ -650,14 +650,14 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
           vtype = T_ARR;
           voff = vstart; vlen = c.pos - vstart;
         } else {
-          {
+          if (depth < SLOT_DEPTH) {
             uint64_t cs = (cur_capture && arr_depth == 0)
                               ? (uint64_t)cur_slot : 31ull;
             cont_slot = (cont_slot & ~(31ull << (5 * depth)))
                         | (cs << (5 * depth));
           }
           if (depth < SIG_DEPTH) sig_stack_at(depth) = cur_sig;
-          is_arr_bits |= (1u << depth);
+          is_arr_bits |= (1ull << depth);
           depth++;
           arr_depth++;
           cur_capture = false; cur_slot = -1;
@@ -711,7 +711,8 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
       if (b == ']') {
         depth--; arr_depth--;
         {
-          int cs = (int)((cont_slot >> (5 * depth)) & 31ull);
+          int cs = (depth < SLOT_DEPTH)
+                       ? (int)((cont_slot >> (5 * depth)) & 31ull) : 31;
           if (cs != 31) fv.set_len(cs, c.pos - fv.get_soff(cs));
         }
         // restore parent sig (not needed for captures inside arrays)
@@ -736,7 +737,8 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
       if (b == '}') {
         depth--;
         {
-          int cs = (int)((cont_slot >> (5 * depth)) & 31ull);
+          int cs = (depth < SLOT_DEPTH)
+                       ? (int)((cont_slot >> (5 * depth)) & 31ull) : 31;
           if (cs != 31) fv.set_len(cs, c.pos - fv.get_soff(cs));
         }
         cur_sig = (depth < SIG_DEPTH) ? sig_stack_at(depth) : 0;
@@ -745,6 +747,107 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
       return false;
     }
   }
+}
+
+// -------------------------------------------------------------------
+// decoded-string compare (cold path: only spans containing '\\').
+// JSON-unescapes the field span on the fly and 3-way-compares it
+// against the constant's raw (unescaped) UTF-8 bytes, mirroring the
+// host-side decode (gpu._decode_json_string) that group keys get.
+// Decoded bytes are packed into a u32 (no local array) so the hot
+// kernel stays scratch-free.
+
+DEV int hexval4(uint8_t x) {
+  if (x >= '0' && x <= '9') return x - '0';
+  if (x >= 'a' && x <= 'f') return x - 'a' + 10;
+  if (x >= 'A' && x <= 'F') return x - 'A' + 10;
+  return -1;
+}
+
+// Decode one logical unit at i (a raw byte or a full escape, \uXXXX
+// surrogate pairs included): packs 1-4 UTF-8 bytes little-endian into
+// out, advances i, returns the byte count (0 = bad escape — cannot
+// happen for spans the tokenizer accepted; defensive).
+DEV int unesc_next(const uint8_t* d, uint32_t fo, uint32_t fl,
+                   uint32_t& i, uint32_t& out) {
+  uint8_t b = d[fo + i];
+  if (b != '\\') { i++; out = b; return 1; }
+  if (i + 2 > fl) return 0;
+  uint8_t e = d[fo + i + 1];
+  i += 2;
+  switch (e) {
+    case '"':  out = '"';  return 1;
+    case '\\': out = '\\'; return 1;
+    case '/':  out = '/';  return 1;
+    case 'b':  out = 8;    return 1;
+    case 'f':  out = 12;   return 1;
+    case 'n':  out = 10;   return 1;
+    case 'r':  out = 13;   return 1;
+    case 't':  out = 9;    return 1;
+    case 'u':  break;
+    default:   return 0;
+  }
+  if (i + 4 > fl) return 0;
+  uint32_t cp = 0;
+  for (int k = 0; k < 4; k++) {
+    int h = hexval4(d[fo + i + k]);
+    if (h < 0) return 0;
+    cp = cp * 16 + (uint32_t)h;
+  }
+  i += 4;
+  if (cp >= 0xD800 && cp < 0xDC00 && i + 6 <= fl &&
+      d[fo + i] == '\\' && d[fo + i + 1] == 'u') {
+    uint32_t lo = 0;
+    bool ok = true;
+    for (int k = 0; k < 4; k++) {
+      int h = hexval4(d[fo + i + 2 + k]);
+      if (h < 0) { ok = false; break; }
+      lo = lo * 16 + (uint32_t)h;
+    }
+    if (ok && lo >= 0xDC00 && lo < 0xE000) {
+      cp = 0x10000 + ((cp - 0xD800) << 10) + (lo - 0xDC00);
+      i += 6;
+    }
+  }
+  if (cp < 0x80) { out = cp; return 1; }
+  if (cp < 0x800) {
+    out = (0xC0 | (cp >> 6)) | ((0x80u | (cp & 63)) << 8);
+    return 2;
+  }
+  if (cp < 0x10000) {
+    out = (0xE0 | (cp >> 12)) | ((0x80u | ((cp >> 6) & 63)) << 8)
+          | ((0x80u | (cp & 63)) << 16);
+    return 3;
+  }
+  out = (0xF0 | (cp >> 18)) | ((0x80u | ((cp >> 12) & 63)) << 8)
+        | ((0x80u | ((cp >> 6) & 63)) << 16)
+        | ((0x80u | (cp & 63)) << 24);
+  return 4;
+}
+
+// 3-way compare of the DECODED field span vs const bytes; -2 on bad
+// escape (defensive — the tokenizer already rejected those records).
+DEV int unesc_cmp(const uint8_t* d, uint32_t fo, uint32_t fl,
+                  const uint8_t* cb, uint32_t cl) {
+  uint32_t i = 0, j = 0;
+  while (i < fl) {
+    uint32_t pack;
+    int n = unesc_next(d, fo, fl, i, pack);
+    if (n == 0) return -2;
+    for (int k = 0; k < n; k++) {
+      uint8_t a = (uint8_t)(pack >> (8 * k));
+      if (j >= cl) return 1;  // field longer than const
+      uint8_t b = cb[j++];
+      if (a != b) return a < b ? -1 : 1;
+    }
+  }
+  return (j == cl) ? 0 : -1;
+}
+
+DEV bool span_has_backslash(const uint8_t* d, uint32_t fo, uint32_t fl) {
+  for (uint32_t k = 0; k < fl; k++)
+    if (d[fo + k] == '\\') return true;
+  return false;
 }
 
 // -------------------------------------------------------------------
@@ -791,6 +894,10 @@ DEV int eval_leaf(const PlanView& P, Bytes BV, const FV& fv,
               if (data[fo + k] != P.const_bytes[co2 + k]) { eq = false; break; }
           }
         }
+        // NON-canonical escapes in the data (GET, \/, surrogate
+        // pairs): unescape-as-you-compare (cold; escaped spans only)
+        if (!eq && span_has_backslash(data, fo, fl))
+          eq = unesc_cmp(data, fo, fl, P.const_bytes + coff, clen) == 0;
       } else {  // string vs number: ToNumber(field)
         double fn = js_to_number(BV, fv.get_soff(slot), fv.get_slen(slot));
         eq = (fn == fn) && (fn == cdval);
@@ -805,12 +912,17 @@ DEV int eval_leaf(const PlanView& P, Bytes BV, const FV& fv,
   if (ft == T_STR && ckind == CONST_STR) {
     uint32_t fo = fv.get_soff(slot), fl = fv.get_slen(slot);
     int cmp = 0;
-    uint32_t n = fl < clen ? fl : clen;
-    for (uint32_t k = 0; k < n; k++) {
-      uint8_t a = data[fo + k], b = P.const_bytes[coff + k];
-      if (a != b) { cmp = a < b ? -1 : 1; break; }
+    if (span_has_backslash(data, fo, fl)) {
+      cmp = unesc_cmp(data, fo, fl, P.const_bytes + coff, clen);
+      if (cmp == -2) cmp = 0;  // unreachable: tokenizer validated
+    } else {
+      uint32_t n = fl < clen ? fl : clen;
+      for (uint32_t k = 0; k < n; k++) {
+        uint8_t a = data[fo + k], b = P.const_bytes[coff + k];
+        if (a != b) { cmp = a < b ? -1 : 1; break; }
+      }
+      if (cmp == 0) cmp = (fl < clen) ? -1 : (fl > clen ? 1 : 0);
     }
-    if (cmp == 0) cmp = (fl < clen) ? -1 : (fl > clen ? 1 : 0);
     switch (op) {
       case OP_LT: return cmp < 0;
       case OP_LE: return cmp <= 0;

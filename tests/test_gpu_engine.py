@@ -227,18 +227,69 @@ def test_escaped_string_filter(engines, tmp_path):
         g = gpu.scan([str(path)], [q])
         cp = c.aggregators[0].points()
         gp = g.aggregators[0].points()
-        # the canonical-escape form matches; \\u000a (non-canonical)
-        # is a documented divergence the CPU oracle does catch
+        # both the canonical escaping (alt-form fast path) and the
+        # non-canonical \\u000a form (unescape-compare fallback) match
         if filt == {"eq": ["m", "a\nb"]}:
             assert cp[0]["value"] == 2
-            assert gp[0]["value"] in (1, 2)  # \\u-escapes divergent
-        else:
-            assert gp == cp
+        assert gp == cp
     # group keys decode identically regardless of escape form
     q = query_load(breakdown_specs="m")
     c = cpu.scan([str(path)], [q])
     g = gpu.scan([str(path)], [q])
     assert g.aggregators[0].points() == c.aggregators[0].points()
+
+
+def test_noncanonical_escape_filters(engines, tmp_path):
+    """Unescape-on-device compare: \\uXXXX-spelled ASCII, \\/, BMP and
+    surrogate-pair escapes, and relational compares over escaped
+    spans all match the CPU oracle exactly."""
+    cpu, gpu = engines
+    from dragnet_amd.query import query_load
+    raw = (b'{"m": "\\u0047ET", "u": "/a\\/b", "s": "x\\u00e9y"}\n'
+           b'{"m": "GET", "u": "/a/b", "s": "x\xc3\xa9y"}\n'
+           b'{"m": "P\\u004fST", "u": "za", "s": "\\ud83d\\ude00"}\n'
+           b'{"m": "GEU", "u": "zb", "s": "\xf0\x9f\x98\x80"}\n')
+    path = tmp_path / "nce.ndjson"
+    path.write_bytes(raw)
+    for filt in [
+        {"eq": ["m", "GET"]},            # GET == GET
+        {"ne": ["m", "GET"]},
+        {"eq": ["u", "/a/b"]},           # \/ == /
+        {"eq": ["s", "xéy"]},       # BMP escape == utf-8 bytes
+        {"eq": ["s", "\U0001F600"]},     # surrogate pair == utf-8
+        {"lt": ["m", "GEU"]},            # relational over escaped span
+        {"ge": ["m", "GET"]},
+    ]:
+        q = query_load(filter=filt)
+        c = cpu.scan([str(path)], [q])
+        g = gpu.scan([str(path)], [q])
+        assert g.aggregators[0].points() == c.aggregators[0].points(), \
+            filt
+
+
+def test_deep_nesting(engines, tmp_path):
+    """Nesting to depth 64 parses identically to the oracle (the
+    capture-slot window is 12 deep, but deeper containers sit inside
+    arrays and are never capture targets)."""
+    cpu, gpu = engines
+    from dragnet_amd.query import query_load
+    lines = []
+    for d in (2, 11, 13, 30, 60):
+        deep = '"leaf"'
+        for i in range(d):
+            deep = ('[%s]' % deep) if i % 2 else ('{"k%d": %s}' % (i, deep))
+        lines.append('{"m": "GET", "deep": %s, "v": %d}' % (deep, d))
+    path = tmp_path / "deep.ndjson"
+    path.write_text("\n".join(lines) + "\n")
+    for spec, filt in [("m", None), ("v", {"eq": ["m", "GET"]}),
+                       (None, {"gt": ["v", 12]})]:
+        q = query_load(filter=filt, breakdown_specs=spec)
+        c = cpu.scan([str(path)], [q])
+        g = gpu.scan([str(path)], [q])
+        assert g.aggregators[0].points() == c.aggregators[0].points(), \
+            (spec, filt)
+        assert dict(g.stages)["json parser"] == \
+            dict(c.stages)["json parser"]
 
 
 def test_overflow_regrow(engines, tmp_path, monkeypatch):

@@ -431,6 +431,13 @@ class _ScanContext(object):
         self._copy_stream = torch.cuda.Stream(device=self.eng.device)
         self._slice_evs = [torch.cuda.Event()
                            for _ in self._slices]
+        # Ping-pong device pools for streaming passes: pass k+1's H2D
+        # lands in the buffer pass k is NOT reading, so the copy
+        # stream runs back-to-back at link rate instead of stalling
+        # behind pass k's kernel tail + extraction + reset.
+        self._dev_data_pp = torch.empty_like(self._dev_data)
+        self._pass_done = [torch.cuda.Event(), torch.cuda.Event()]
+        self._db_idx = 0
 
         # Pre-index line positions per slice: the pool bytes are
         # identical on every pass, so the line index is computed once
@@ -457,8 +464,15 @@ class _ScanContext(object):
         pin = self._pinned
         main = torch.cuda.current_stream(self.eng.device)
         if h2d:
-            # enqueue all slice copies on the copy stream
-            self._copy_stream.wait_stream(main)  # table reset ordering
+            # ping-pong: write the buffer the PREVIOUS pass did not
+            # read; wait only for the pass-before-last (the last
+            # reader of this buffer), which has long finished — so
+            # the copies chain back-to-back across passes
+            idx = self._db_idx
+            self._db_idx ^= 1
+            dev_data = self._dev_data if idx == 0 else self._dev_data_pp
+            ev_done = self._pass_done[idx]
+            self._copy_stream.wait_event(ev_done)
             with torch.cuda.stream(self._copy_stream):
                 for k, (s, e) in enumerate(self._slices):
                     s16 = s & ~15
@@ -480,6 +494,8 @@ class _ScanContext(object):
             else:
                 pos_k, nlines_k = self._slice_pos[k]
                 self._scan_call(dev_data, s, pos_k, nlines_k)
+        if h2d:
+            ev_done.record(main)  # last reader of this pass's buffer
 
     def _scan_call(self, dev_data, first_start, pos=None, nlines=None):
         self.eng.ops.scan_chunk(

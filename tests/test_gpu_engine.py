@@ -267,6 +267,47 @@ def test_noncanonical_escape_filters(engines, tmp_path):
             filt
 
 
+def test_resident_streaming_passes(engines, tmp_path):
+    """The bench's streaming path: stage a pool once, run several
+    full passes (ping-pong device pools, sliced copy/kernel overlap,
+    pipelined extraction) — every pass must equal the oracle, and
+    pipelined decode of pass k must survive pass k+1 running."""
+    cpu, gpu = engines
+    from dragnet_amd.engine import plan as planmod
+    from dragnet_amd.engine.gpu import _ScanContext
+    from dragnet_amd.query import query_load
+    from dragnet_amd.tools.mktestdata import generate_lines
+    lines = []
+    for line in generate_lines(20000, seed=77):
+        lines.append(line)
+    pool = b"".join(lines)
+    path = tmp_path / "pool.ndjson"
+    path.write_bytes(pool)
+    q = query_load(filter={"eq": ["req.method", "GET"]},
+                   breakdown_specs="req.method,res.statusCode")
+    expected = cpu.scan([str(path)], [q]).aggregators[0].points()
+
+    cplan = planmod.compile_plan([q])
+    ctx = _ScanContext(gpu, cplan, agg_slots=1 << 14,
+                       dict_slots=1 << 14, dict_data_cap=4 << 20)
+    ctx.stage_resident(pool)
+    prev = None
+    results = []
+    for _ in range(4):
+        ctx.reset()
+        ctx.scan_resident(h2d=True)
+        ex = ctx.extract_async([q])
+        if prev is not None:
+            aggs, _ = ctx.decode_extracted(prev, [q])
+            results.append(aggs[0].points())
+        prev = ex
+    aggs, _ = ctx.decode_extracted(prev, [q])
+    results.append(aggs[0].points())
+    assert len(results) == 4
+    for r in results:
+        assert r == expected
+
+
 def test_deep_nesting(engines, tmp_path):
     """Nesting to depth 64 parses identically to the oracle (the
     capture-slot window is 12 deep, but deeper containers sit inside

@@ -124,7 +124,9 @@ def merge_tables_tensor(agg, query, device, group=None):
              for _ in range(dist.get_world_size(group))]
     mine = torch.tensor([t.numel()], dtype=torch.int64, device=device)
     dist.all_gather(sizes, mine, group=group)
-    maxn = int(max(s.item() for s in sizes))
+    # >=1 so zero-length all_gather never reaches RCCL (all-empty
+    # tables, e.g. a filter matching nothing anywhere)
+    maxn = max(int(max(s.item() for s in sizes)), 1)
     padded = torch.zeros(maxn, dtype=torch.uint8, device=device)
     padded[:t.numel()] = t
     bufs = [torch.zeros(maxn, dtype=torch.uint8, device=device)

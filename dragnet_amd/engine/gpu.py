@@ -118,6 +118,7 @@ class _ScanContext(object):
         # plan buffers
         self.field_sigs = torch.from_numpy(
             cplan.field_sigs.view(np.int64)).to(dev)
+        self.comp_slot = torch.from_numpy(cplan.comp_slot).to(dev)
         progs, bounds = cplan.programs
         self.prog_nodes = dev_i32(progs)
         self.prog_bounds = dev_i32(bounds)
@@ -503,7 +504,9 @@ class _ScanContext(object):
             pos if pos is not None else self._pos,
             nlines if nlines is not None else self._nlines,
             first_start,
-            self.field_sigs, self.prog_nodes, self.prog_bounds,
+            self.field_sigs, self.comp_slot, self.cplan.nf_match,
+            self.cplan.sig_bloom,
+            self.prog_nodes, self.prog_bounds,
             self.const_meta, self.const_dvals, self.const_bytes,
             self.synth_slots, self.cplan.n_synth,
             self.metric_rows, self.synth_req,

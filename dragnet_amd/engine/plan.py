@@ -21,6 +21,8 @@ Layouts are mirrored by struct definitions in ops/hip/common.h.
 """
 
 
+import os
+
 import numpy as np
 
 from .. import krill
@@ -77,6 +79,12 @@ def comp_hash(b):
         w = int.from_bytes(b[k:k + 8].ljust(8, b"\0"), "little")
         h = _mix64(h ^ w)
     return _mix64(h ^ len(b))
+
+
+def lit_sig(name):
+    """Signature of a whole literal key (dots as plain bytes) at the
+    top level — the companion-slot match target for json mode."""
+    return _mix64(FNV_OFFSET ^ comp_hash(name.encode("utf-8")))
 
 
 def path_sig(path):
@@ -328,6 +336,30 @@ def compile_plan(queries, ds_filter=None, time_field=None,
         value_slot = fields.slot("value", raw=True)
         fields_slot = fields.slot("fields", raw=True)
 
+    # Companion slots: a TOP-LEVEL literal dotted key ("a.b": v) is
+    # addressable by the aggregation lookup (points.lookup checks the
+    # whole literal name before plucking) but invisible to krill pluck
+    # and synthetic sources.  In json mode the device hashes each key
+    # as ONE component (dots are plain bytes), so the companion's sig
+    # below — the whole literal name as one root-level component —
+    # captures exactly those keys into the companion physical slot;
+    # breakdown key extraction reads it first, everything else reads
+    # the primary (pluck) slot.
+    nf_match = len(fields.paths)
+    comp_slot = np.full(max(nf_match, 1), -1, dtype=np.int32)
+    comp_sigs = []
+    if data_format != "json-skinner" and \
+            os.environ.get("DRAGNET_NO_COMPANIONS") != "1":
+        for i, (pth, raw) in enumerate(fields.paths):
+            if not raw and "." in pth:
+                phys = nf_match + len(comp_sigs)
+                if phys >= MAX_FIELDS:
+                    raise PlanError(
+                        "query references more than %d fields "
+                        "(incl. dotted-key companions)" % MAX_FIELDS)
+                comp_slot[i] = phys
+                comp_sigs.append(lit_sig(pth))
+
     programs = np.array(prog_nodes, dtype=np.int32).reshape(-1, 4)
     bounds = np.array(prog_bounds, dtype=np.int32).reshape(-1, 2)
     const_meta = np.array(consts.metas, dtype=np.int32).reshape(-1, 6)
@@ -345,6 +377,17 @@ def compile_plan(queries, ds_filter=None, time_field=None,
         fields, (programs, bounds), const_meta, const_dvals,
         const_bytes, synthetic, (metrics, sreq), (bds, steps), queries)
     plan.n_synth = len(synth_slots)
+    plan.comp_slot = comp_slot
+    plan.nf_match = nf_match
+    bloom = 0
+    for sg in ([int(x) for x in fields.sigs()] + comp_sigs):
+        bloom |= 1 << (sg & 63)
+    # int64 for the binding (torch scalar args are signed)
+    plan.sig_bloom = bloom - (1 << 64) if bloom >= (1 << 63) else bloom
+    if comp_sigs:
+        plan.field_sigs = np.concatenate(
+            [plan.field_sigs,
+             np.array(comp_sigs, dtype=np.uint64)])
     plan.value_slot = value_slot
     plan.fields_slot = fields_slot
     plan.ds_prog = ds_prog

@@ -111,8 +111,13 @@ struct NumDict {
 // ---- kernel argument structs (shared between kernels and bindings) ----
 
 struct PlanView {
-  const uint64_t* field_sigs;   // [nf]
-  int nf;
+  const uint64_t* field_sigs;   // [nf] (companions carry poison sigs)
+  int nf;                       // total physical slots (LDS sizing)
+  const int32_t* comp_slot;     // [nf_match] literal-dotted companion
+                                // slot per primary field, or -1
+  int nf_match;                 // primary field count (agg readout)
+  uint64_t sig_bloom;           // OR of 1<<(sig&63): cheap pre-filter
+                                // so non-matching keys skip the loop
   const int32_t* prog_nodes;    // [n_nodes][4]
   const int32_t* prog_bounds;   // [n_progs][2]
   const int32_t* const_meta;    // [nc][6] kind, off, len, dvalid, off2, len2

@@ -98,7 +98,8 @@ void newline_index(torch::Tensor data, int64_t start, int64_t n,
 void scan_chunk(
     torch::Tensor data, torch::Tensor nl_pos, torch::Tensor nlines_dev,
     int64_t first_start,
-    torch::Tensor field_sigs, torch::Tensor prog_nodes,
+    torch::Tensor field_sigs, torch::Tensor comp_slot,
+    int64_t nf_match, int64_t sig_bloom, torch::Tensor prog_nodes,
     torch::Tensor prog_bounds, torch::Tensor const_meta,
     torch::Tensor const_dvals, torch::Tensor const_bytes,
     torch::Tensor synth_slots, int64_t n_synth,
@@ -126,6 +127,9 @@ void scan_chunk(
 
   A.P.field_sigs = (const uint64_t*)field_sigs.data_ptr();
   A.P.nf = (int)field_sigs.numel();
+  A.P.comp_slot = (const int32_t*)comp_slot.data_ptr();
+  A.P.nf_match = (int)nf_match;
+  A.P.sig_bloom = (uint64_t)sig_bloom;
   A.P.prog_nodes = (const int32_t*)prog_nodes.data_ptr();
   A.P.prog_bounds = (const int32_t*)prog_bounds.data_ptr();
   A.P.const_meta = (const int32_t*)const_meta.data_ptr();

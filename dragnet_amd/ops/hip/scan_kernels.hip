@@ -237,10 +237,19 @@ DEV uint64_t comp_hash_span(Bytes B, uint32_t s, uint32_t e) {
 
 // Scan a KEY (cursor after the opening quote): SWAR windows to the
 // next quote/dot/escape/control; chains component hashes into the
-// path signature (mirrors plan.path_sig — '.' splits components so
-// literal dotted keys unify with nested paths).  Escaped keys are
-// validated but get a sentinel signature that matches no field.
-DEV int scan_key_sig(Cursor& c, uint64_t parent, uint64_t& sig_out) {
+// path signature (mirrors plan.path_sig).  dot_splits is the mode
+// switch: json-skinner reads point.fields[name] literally with
+// dotted column names, so in-key '.' chains components there; json
+// mode hashes the key as ONE component (dots are plain bytes), so a
+// nested path matches its chained primary sig while a TOP-LEVEL
+// literal "a.b" key matches the field's companion sig (the whole
+// literal name as one component — plan.py lit_sig), which the
+// aggregation readout consults first (points.lookup is literal-first)
+// and krill pluck / synthetic sources never see.  Keys nested UNDER a
+// literal dotted key chain from the literal sig and match nothing.
+// Escaped keys are validated but get a sentinel signature.
+DEV int scan_key_sig(Cursor& c, uint64_t parent, uint64_t& sig_out,
+                     bool dot_splits) {
   const uint8_t* d = c.B.mem - c.B.bias;
   uint32_t p = c.pos, end = c.end;
   uint32_t comp_s = p;
@@ -263,9 +272,11 @@ DEV int scan_key_sig(Cursor& c, uint64_t parent, uint64_t& sig_out) {
       return 1;
     }
     if (b == '.') {
-      sig = mix64(sig ^ comp_hash_span(c.B, comp_s, p));
+      if (dot_splits) {
+        sig = mix64(sig ^ comp_hash_span(c.B, comp_s, p));
+        comp_s = p + 1;
+      }
       p++;
-      comp_s = p;
       continue;
     }
     if (b == '\\') {
@@ -555,7 +566,7 @@ DEV DateOut parse_iso_ms(Bytes BV, uint32_t off, uint32_t len) {
 // JSON.  top_type receives the top-level value type.
 DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
                       const PlanView& P, FV& fv, uint8_t& top_type,
-                      uint64_t* sig_lds) {
+                      uint64_t* sig_lds, bool dot_splits) {
   Cursor c; c.init(BV, start, end);
   // parent-signature stack in LDS: [depth * BLOCK + tid]
   #define sig_stack_at(d) sig_lds[(d) * BLOCK + (uint32_t)threadIdx.x]
@@ -573,6 +584,7 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
 
   // match cur sig against the field table
   auto match_slot = [&](uint64_t sig) -> int {
+    if (!((P.sig_bloom >> ((uint32_t)sig & 63u)) & 1ull)) return -1;
     for (int f = 0; f < P.nf; f++)
       if (P.field_sigs[f] == sig) return f;
     return -1;
@@ -584,7 +596,8 @@ DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
   // parent signature via chained component hashes (plan.path_sig)
   auto parse_key = [&](uint64_t parent, bool root, uint64_t& sig_out) -> bool {
     if (c.eof() || c.next() != '"') return false;
-    return scan_key_sig(c, root ? FNV_OFFSET : parent, sig_out) != 0;
+    return scan_key_sig(c, root ? FNV_OFFSET : parent, sig_out,
+                        dot_splits) != 0;
   };
 
   // Main loop: parse values iteratively.
@@ -1358,7 +1371,8 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
       uint32_t end = A.nl_pos[r];
       uint8_t top_type;
       bool ok = (end > start) &&
-                parse_record(BV, start, end, P, fv, top_type, sig_lds);
+                parse_record(BV, start, end, P, fv, top_type, sig_lds,
+                             A.data_format_skinner);
       double weight = 1.0;
       if (ok && A.data_format_skinner) {
         // require: object top, a "fields" member, numeric "value"
@@ -1450,6 +1464,12 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
                 } else t = T_MISSING;  // cannot happen: required above
               } else {
                 int slot = B[1];
+                // aggregation lookup is literal-first (points.lookup):
+                // a top-level literal dotted key (companion slot)
+                // shadows the plucked nested value
+                int cs2 = P.comp_slot[slot];
+                if (cs2 >= 0 && fv.get_type(cs2) != T_MISSING)
+                  slot = cs2;
                 t = fv.get_type(slot);
                 num = fv.get_num(slot);
                 so = fv.get_soff(slot); sl = fv.get_slen(slot);

@@ -308,6 +308,32 @@ def test_resident_streaming_passes(engines, tmp_path):
         assert r == expected
 
 
+def test_literal_dotted_keys(engines, tmp_path):
+    """Reference lookup split: krill pluck / synthetic sources never
+    see a literal "a.b" key, but the aggregation lookup is
+    literal-first at the top level (points.lookup).  The kernel routes
+    literal-dotted captures to companion slots."""
+    cpu, gpu = engines
+    from dragnet_amd.query import query_load
+    raw = (b'{"a.b": 1, "m": "w"}\n'
+           b'{"a": {"b": 2}, "m": "x"}\n'
+           b'{"a.b": 3, "a": {"b": 4}, "m": "y"}\n'
+           b'{"x": {"a.b": 5}, "m": "z"}\n'
+           b'{"a.b": {"c": 6}, "m": "q"}\n')
+    path = tmp_path / "dot.ndjson"
+    path.write_bytes(raw)
+    for spec, filt in [("a.b", None), ("x.a.b", None), ("a.b.c", None),
+                       ("m", {"eq": ["a.b", 2]}),
+                       (None, {"gt": ["a.b", 0]}),
+                       ("a.b,m", {"eq": ["m", "y"]})]:
+        q = query_load(filter=filt, breakdown_specs=spec)
+        c = cpu.scan([str(path)], [q])
+        g = gpu.scan([str(path)], [q])
+        assert g.aggregators[0].points() == c.aggregators[0].points(), \
+            (spec, filt)
+        assert g.aggregators[0].ninputs == c.aggregators[0].ninputs
+
+
 def test_deep_nesting(engines, tmp_path):
     """Nesting to depth 64 parses identically to the oracle (the
     capture-slot window is 12 deep, but deeper containers sit inside

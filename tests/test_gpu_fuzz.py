@@ -18,7 +18,8 @@ pytestmark = pytest.mark.gpu
 # NOTE: literal dotted keys (e.g. "x.y") are excluded: the GPU's
 # path-signature lookup unifies them with nested paths while krill's
 # eval plucks only nested paths (documented divergence, COMPONENTS.md)
-KEYS = ["a", "b", "c", "req", "res", "xy", "time", "lat", "msg"]
+KEYS = ["a", "b", "c", "req", "res", "xy", "time", "lat", "msg",
+        "a.b", "req.a"]  # literal dotted keys: companion-slot path
 STRINGS = ["GET", "PUT", "", "hello world", "héllo", "line\nbreak",
            'quo"te', "back\\slash", "tab\there", "200", "1e3", "0x10",
            "  12 ", "Infinity", "naan", "ünïcødé-αβγ", "a" * 120]

@@ -79,7 +79,11 @@ def main(argv=None):
         track_time = True
         argv.pop(0)
 
+    from .log import get_logger
+    log = get_logger()
+    log.debug("dispatch", argv=argv)
     rv = _dispatch(argv)
+    log.debug("done", exit_status=rv)
     if track_time:
         total = time.monotonic() - t_start
         sys.stderr.write("timing stats:\n")

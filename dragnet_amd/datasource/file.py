@@ -346,6 +346,10 @@ def find_files(roots):
 def write_index(index_path, metrics, interval, points):
     """Route aggregated points into per-interval IndexSinks; atomic
     rename on flush.  Returns index file paths written."""
+    from ..log import get_logger
+    _log = get_logger().child("datasource-file")
+    _log.info("writing index", index_path=index_path,
+              interval=interval, nmetrics=len(metrics))
     written = []
     if interval == "all":
         sink = IndexSink(os.path.join(index_path, "all"), metrics)

@@ -87,6 +87,10 @@ class GpuEngine(object):
             else:
                 ctx.scan_files(files)
             if ctx.overflowed():
+                from ..log import get_logger
+                get_logger().child("gpu-engine").warn(
+                    "aggregation table overflow; restarting scan",
+                    agg_slots=agg_slots * 8, dict_slots=dict_slots * 8)
                 agg_slots *= 8
                 dict_slots *= 8
                 dict_data *= 4

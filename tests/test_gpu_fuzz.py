@@ -7,7 +7,7 @@ significant digits, ASCII keys, nesting within the validated depth.
 Everything else is fair game: missing fields, nulls, type mixes,
 malformed/truncated lines, empty lines, duplicate keys, literal
 dotted keys (companion slots), unicode values in BOTH canonical and
-\uXXXX-escaped spellings (incl. surrogate pairs), deep nesting.
+backslash-u-escaped spellings (incl. surrogate pairs), deep nesting.
 """
 
 import json
@@ -62,7 +62,7 @@ def rand_line(rng):
         return json.dumps(rec_or_scalar(rng),
                           ensure_ascii=False).encode()
     if r < 0.85:
-        # \uXXXX-escaped spelling of the same value space (keys stay
+        # backslash-u-escaped spelling of the same values (keys stay
         # ASCII; escaped keys are the remaining documented divergence)
         return json.dumps(rec_or_scalar(rng),
                           ensure_ascii=True).encode()

@@ -207,6 +207,28 @@ void scan_chunk(
               hipGetErrorString(err));
 }
 
+// One-call zeroing of all scan state: a streaming step otherwise
+// issues ~10 separate torch .zero_() dispatches whose python+dispatch
+// overhead shows at 5 ms/step.
+void scan_reset(std::vector<torch::Tensor> states,
+                std::vector<torch::Tensor> counts,
+                torch::Tensor sd_state, torch::Tensor sd_used,
+                torch::Tensor sd_next, torch::Tensor nd_state,
+                torch::Tensor nd_next, torch::Tensor counters) {
+  hipStream_t st = current_stream();
+  auto z = [&](torch::Tensor& t) {
+    hipError_t e = hipMemsetAsync(t.data_ptr(), 0,
+                                  (size_t)t.numel() * t.element_size(),
+                                  st);
+    TORCH_CHECK(e == hipSuccess, "scan_reset memset failed: ",
+                hipGetErrorString(e));
+  };
+  for (auto& t : states) z(t);
+  for (auto& t : counts) z(t);
+  z(sd_state); z(sd_used); z(sd_next);
+  z(nd_state); z(nd_next); z(counters);
+}
+
 // Build the device-side AggTable descriptor array from per-metric
 // tensors.  Returns a CPU byte tensor; the caller copies it to the GPU.
 torch::Tensor agg_descs_host(std::vector<torch::Tensor> states,
@@ -311,6 +333,42 @@ torch::Tensor extract_numdict(torch::Tensor nd_state,
   return out;
 }
 
+// One-call snapshot + extraction for the pipelined finalize
+// (gpu._ScanContext.extract_async): clones of the counters and
+// dictionary cursors/blob, the dictionary extractions, and the
+// no-sync per-metric aggregate extraction, enqueued on the current
+// stream in one binding call instead of ~10 python dispatches.
+// Returns [cnt, n_str, n_num, used, blob, str_off, str_len, numbers,
+// then (keys, counts, n) per metric].
+std::vector<torch::Tensor> extract_all(
+    std::vector<torch::Tensor> states, std::vector<torch::Tensor> keys,
+    std::vector<torch::Tensor> counts, int64_t max_out,
+    torch::Tensor sd_state, torch::Tensor sd_hash, torch::Tensor sd_id,
+    torch::Tensor sd_off, torch::Tensor sd_len, torch::Tensor sd_data,
+    torch::Tensor sd_used, torch::Tensor sd_next,
+    torch::Tensor nd_state, torch::Tensor nd_bits, torch::Tensor nd_id,
+    torch::Tensor nd_next, torch::Tensor counters, int64_t dict_slots) {
+  std::vector<torch::Tensor> out;
+  out.push_back(counters.clone());
+  out.push_back(sd_next.clone());
+  out.push_back(nd_next.clone());
+  out.push_back(sd_used.clone());
+  // the dictionary byte blob is bump-allocated from 0 each scan, so
+  // the NEXT step overwrites it — snapshot now
+  out.push_back(sd_data.clone());
+  auto so = extract_strdict(sd_state, sd_hash, sd_id, sd_off, sd_len,
+                            sd_data, sd_used, sd_next, dict_slots);
+  out.push_back(so[0]);
+  out.push_back(so[1]);
+  out.push_back(extract_numdict(nd_state, nd_bits, nd_id, nd_next,
+                                dict_slots));
+  for (size_t m = 0; m < states.size(); m++) {
+    auto a = extract_agg_async(states[m], keys[m], counts[m], max_out);
+    out.insert(out.end(), a.begin(), a.end());
+  }
+  return out;
+}
+
 }  // namespace dn
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -320,6 +378,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("agg_descs_host", &dn::agg_descs_host);
   m.def("extract_agg", &dn::extract_agg);
   m.def("extract_agg_async", &dn::extract_agg_async);
+  m.def("extract_all", &dn::extract_all);
+  m.def("scan_reset", &dn::scan_reset);
   m.def("extract_strdict", &dn::extract_strdict);
   m.def("extract_numdict", &dn::extract_numdict);
   m.attr("MAX_KEY") = (int)dn::MAX_KEY;

@@ -262,7 +262,7 @@ class _ScanContext(object):
         views = [memoryview(p.numpy()) for p in pins]
         evs = [None, None]
         pool = cf.ThreadPoolExecutor(
-            max_workers=_env_int("DRAGNET_READERS", 4))
+            max_workers=_env_int("DRAGNET_READERS", 8))
 
         def pread_full(fd, mv, off):
             """preadv until mv is full (a single preadv may legally
@@ -288,7 +288,7 @@ class _ScanContext(object):
                 return seq_file.readinto(view[at:at + want]) or 0
             if want < (8 << 20):
                 return pread_full(fd, view[at:at + want], fpos)
-            nsec = _env_int("DRAGNET_READERS", 4)
+            nsec = _env_int("DRAGNET_READERS", 8)
             sec = (want + nsec - 1) // nsec
             futs = []
             for s in range(0, want, sec):

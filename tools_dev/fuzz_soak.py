@@ -2,14 +2,15 @@
 import sys
 sys.path.insert(0, "/root/repo")
 sys.path.insert(0, "/root/repo/tests")
-import random, tempfile, os
+import os, random, tempfile
 from test_gpu_fuzz import rand_line, rand_query
 from dragnet_amd.engine.cpu import CpuEngine
 from dragnet_amd.engine.gpu import GpuEngine
 
 cpu, gpu = CpuEngine(), GpuEngine()
 fails = 0
-for seed in range(100, 132):
+N = int(os.environ.get("SOAK_SEEDS", "32"))
+for seed in range(100, 100 + N):
     rng = random.Random(seed)
     lines = [rand_line(rng) for _ in range(1500)]
     with tempfile.NamedTemporaryFile(suffix=".ndjson", delete=False) as f:
@@ -27,5 +28,5 @@ for seed in range(100, 132):
                          [b["name"] for b in q.breakdowns]))
     finally:
         os.unlink(path)
-print("soak done: %d divergences over 32 seeds x 5 queries" % fails)
+print("soak done: %d divergences over %d seeds x 5 queries" % (fails, N))
 assert fails == 0

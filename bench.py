@@ -114,7 +114,12 @@ def main():
     ctx.stage_resident(pool)
     device = eng.device
 
-    if args.device_resident:
+    xpose = (args.device_resident
+             and os.environ.get("DRAGNET_XPOSE") == "1")
+    if xpose:
+        log("staging wave-transposed pool ...")
+        ctx.stage_xpose(pool)
+    if args.device_resident and not xpose:
         ctx.scan_resident()  # prime HBM once
 
     # hipGraph replay pays on the device-resident path (one launch per
@@ -122,7 +127,8 @@ def main():
     # and HURT the streaming path (35 vs 47 GB/s measured), so graphs
     # stay off when copies are in the loop.
     graph = None
-    if args.device_resident and os.environ.get("DRAGNET_NO_GRAPH") != "1":
+    if (args.device_resident and not xpose
+            and os.environ.get("DRAGNET_NO_GRAPH") != "1"):
         graph = ctx.make_graph(h2d=False)
         log("hipGraph capture: %s"
             % ("ok" if graph is not None else "unavailable"))
@@ -145,6 +151,9 @@ def main():
     def step():
         if graph is not None:
             graph.replay()
+        elif xpose:
+            ctx.reset()
+            ctx.scan_xpose()
         else:
             ctx.reset()
             ctx.scan_resident(h2d=not args.device_resident)
@@ -217,8 +226,10 @@ def main():
             "dtype": "uint8",
             "data": "synthetic (mktestdata muskie-shaped NDJSON, "
                     "random-generated, %d MB/GPU %s pool)"
-                    % (args.mb, "device-resident"
-                       if args.device_resident else "host-staged"),
+                    % (args.mb,
+                       "device-resident wave-transposed" if xpose
+                       else "device-resident" if args.device_resident
+                       else "host-staged"),
             "gb_per_sec": round(gb_per_sec, 3),
             "config": {
                 "model": "dn scan: filter eq(req.method,GET) + "

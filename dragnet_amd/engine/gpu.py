@@ -502,7 +502,8 @@ class _ScanContext(object):
         if h2d:
             ev_done.record(main)  # last reader of this pass's buffer
 
-    def _scan_call(self, dev_data, first_start, pos=None, nlines=None):
+    def _scan_call(self, dev_data, first_start, pos=None, nlines=None,
+                   x=None):
         self.eng.ops.scan_chunk(
             dev_data,
             pos if pos is not None else self._pos,
@@ -524,7 +525,71 @@ class _ScanContext(object):
             self.sd["used"], self.sd["next"],
             self.nd["state"], self.nd["bits"], self.nd["id"],
             self.nd["next"],
-            self.counters)
+            self.counters,
+            x["xdata"] if x else dev_data,
+            x["wave_base"] if x else self.counters,
+            x["rec_len"] if x else self.counters,
+            x["n_slots"] if x else 0)
+
+    # ---- wave-transposed staging (prototype; DRAGNET_XPOSE) ----
+
+    def stage_xpose(self, buf):
+        """Re-lay a byte pool wave-transposed for scan_kernel_x:
+        records length-sorted (aggregation is order-independent), 64
+        consecutive sorted records form a wave, and each record's
+        bytes are split into 64B granules interleaved so granule g of
+        lane l sits at wave_base + g*4096 + l*64 — a wave's window
+        refills then touch 64 CONSECUTIVE granules (coalesced) instead
+        of 64 scattered records (the SQ_WAIT-bound gather profile in
+        profiles/).  Prototype uses one global granule count (max
+        record length), fine for near-uniform record sizes."""
+        torch = self.t
+        dev = self.eng.device
+        arr = np.frombuffer(buf, dtype=np.uint8)
+        nl = np.flatnonzero(arr == 10).astype(np.int64)
+        if nl.size == 0:
+            raise ValueError("no records")
+        starts = np.empty_like(nl)
+        starts[0] = 0
+        starts[1:] = nl[:-1] + 1
+        lens = (nl - starts).astype(np.int64)
+        n = int(lens.size)
+        order = np.argsort(lens, kind="stable")
+        nslots = (n + 63) & ~63
+        g = max(1, (int(lens.max()) + 63) // 64)
+        K = g * 64
+        # gather each sorted record into a dense (n, K) matrix
+        so = starts[order]
+        lo = lens[order]
+        idx = so[:, None] + np.arange(K)[None, :]
+        np.minimum(idx, arr.size - 1, out=idx)
+        m = arr[idx]
+        m[np.arange(K)[None, :] >= lo[:, None]] = 10
+        if nslots > n:
+            m = np.concatenate(
+                [m, np.full((nslots - n, K), 10, dtype=np.uint8)])
+        nw = nslots // 64
+        # (wave, lane, granule, 64) -> (wave, granule, lane, 64)
+        xb = np.ascontiguousarray(
+            m.reshape(nw, 64, g, 64).transpose(0, 2, 1, 3)).reshape(-1)
+        xb = np.concatenate([xb, np.full(4096, 10, dtype=np.uint8)])
+        slot_len = np.full(nslots, 0xFFFFFFFF, dtype=np.uint32)
+        slot_len[:n] = lo.astype(np.uint32)
+        wave_base = (np.arange(nw, dtype=np.int64) * (g * 4096))
+        self._x = {
+            "xdata": torch.from_numpy(xb).to(dev),
+            "wave_base": torch.from_numpy(wave_base).to(dev),
+            "rec_len": torch.from_numpy(
+                slot_len.view(np.int32)).to(dev),
+            "n_slots": nslots,
+        }
+        self._x_nrec = n
+
+    def scan_xpose(self):
+        """One scan pass over the wave-transposed staging."""
+        rl = self._x["rec_len"]
+        self._scan_call(self._x["xdata"], 0, pos=rl, nlines=rl,
+                        x=self._x)
 
     def make_graph(self, h2d=True):
         """Capture reset + the whole sliced scan pass into a hipGraph

@@ -147,6 +147,11 @@ struct ScanArgs {
   NumDict ndict;
   unsigned long long* counters;
   int data_format_skinner;
+  // wave-transposed staging (scan_kernel_x; null for linear scans)
+  const uint8_t* xdata;              // transposed pool
+  const unsigned long long* xwave_base;  // [nwaves] byte base per wave
+  const uint32_t* xrec_len;          // [n_slots] record len; ~0u = pad
+  uint32_t xn_slots;                 // slots = records rounded to 64
 };
 
 }  // namespace dn

@@ -113,7 +113,10 @@ void scan_chunk(
     torch::Tensor sd_used, torch::Tensor sd_next,
     torch::Tensor nd_state, torch::Tensor nd_bits, torch::Tensor nd_id,
     torch::Tensor nd_next,
-    torch::Tensor counters) {
+    torch::Tensor counters,
+    // wave-transposed staging (xn_slots > 0 selects scan_kernel_x)
+    torch::Tensor xdata, torch::Tensor xwave_base,
+    torch::Tensor xrec_len, int64_t xn_slots) {
   CHECK_GPU(data);
   CHECK_GPU(nl_pos);
   CHECK_GPU(table_descs);
@@ -151,6 +154,15 @@ void scan_chunk(
   A.ndict = make_ndict(nd_state, nd_bits, nd_id, nd_next);
   A.counters = (unsigned long long*)counters.data_ptr();
   A.data_format_skinner = skinner ? 1 : 0;
+  A.xdata = nullptr; A.xwave_base = nullptr; A.xrec_len = nullptr;
+  A.xn_slots = 0;
+  if (xn_slots > 0) {
+    CHECK_GPU(xdata);
+    A.xdata = (const uint8_t*)xdata.data_ptr();
+    A.xwave_base = (const unsigned long long*)xwave_base.data_ptr();
+    A.xrec_len = (const uint32_t*)xrec_len.data_ptr();
+    A.xn_slots = (uint32_t)xn_slots;
+  }
 
   int nf = A.P.nf;
   size_t lds = (size_t)nf * BLOCK * (8 + 4 + 4 + 1);
@@ -184,6 +196,23 @@ void scan_chunk(
   // register-allocator cap measured +17% over the unconstrained build
   const char* mw_env = getenv("DRAGNET_MIN_WAVES");
   int mw = mw_env ? atoi(mw_env) : 4;
+  if (A.xn_slots > 0) {
+    blocks = (A.xn_slots + BLOCK - 1) / BLOCK;
+    if (blocks > 2048) blocks = 2048;
+    if (mw == 2)
+      hipLaunchKernelGGL(scan_kernel_x<2>, dim3(blocks), dim3(BLOCK),
+                         lds, current_stream(), A);
+    else if (mw == 3)
+      hipLaunchKernelGGL(scan_kernel_x<3>, dim3(blocks), dim3(BLOCK),
+                         lds, current_stream(), A);
+    else
+      hipLaunchKernelGGL(scan_kernel_x<4>, dim3(blocks), dim3(BLOCK),
+                         lds, current_stream(), A);
+    hipError_t xerr = hipGetLastError();
+    TORCH_CHECK(xerr == hipSuccess, "scan_kernel_x launch failed: ",
+                hipGetErrorString(xerr));
+    return;
+  }
   if (mw == 2)
     hipLaunchKernelGGL(scan_kernel_mw<2>, dim3(blocks), dim3(BLOCK),
                        lds, current_stream(), A);

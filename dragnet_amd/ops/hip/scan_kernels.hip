@@ -113,19 +113,59 @@ struct Bytes {
   uint32_t bias;
   DEV uint8_t at(uint32_t abs) const { return mem[abs - bias]; }
   DEV const uint8_t* ptr(uint32_t abs) const { return mem + (abs - bias); }
+  // 8 little-endian bytes starting at abs (may read 7 past a span;
+  // host padding guarantees slack)
+  DEV uint64_t load8(uint32_t abs) const {
+    uint64_t w;
+    __builtin_memcpy(&w, mem + (abs - bias), 8);
+    return w;
+  }
 };
 
-struct Cursor {
-  Bytes B;
+// Wave-transposed byte source (DRAGNET_XPOSE resident staging): the
+// pool is re-laid out so granule g of the record owned by lane l of a
+// wave sits at wave_base + g*(64*XGRAN) + l*XGRAN.  A wave's 64
+// window refills for granule g then touch 64 *consecutive* XGRAN-byte
+// blocks — fully coalesced, ~4x fewer cache lines than the linear
+// layout's 64 scattered records (the SQ_WAIT-bound gather profile).
+// Positions are RECORD-RELATIVE (start=0); spans captured from this
+// source are decoded through the same XBytes.
+constexpr uint32_t XGRAN = 64;            // granule bytes
+constexpr uint32_t XSTRIDE = 64 * XGRAN;  // wave bytes per granule row
+
+struct XBytes {
+  const uint8_t* base;  // pool + wave_base + lane*XGRAN
+  DEV uint8_t at(uint32_t p) const {
+    return base[(size_t)(p >> 6) * XSTRIDE + (p & 63u)];
+  }
+  DEV uint64_t load8(uint32_t p) const {
+    uint32_t o = p & 63u;
+    const uint8_t* gp = base + (size_t)(p >> 6) * XSTRIDE;
+    uint64_t w;
+    if (o <= XGRAN - 8) {
+      __builtin_memcpy(&w, gp + o, 8);
+      return w;
+    }
+    uint64_t lo, hi;
+    __builtin_memcpy(&lo, gp + XGRAN - 8, 8);
+    __builtin_memcpy(&hi, gp + XSTRIDE, 8);
+    uint32_t sh = (o - (XGRAN - 8)) * 8;  // 8..56
+    return (lo >> sh) | (hi << (64 - sh));
+  }
+};
+
+template <class BS>
+struct CursorT {
+  BS B;
   uint32_t pos, end;
   uint64_t win, win2;  // 16 record bytes at [wbase, wbase+16)
   uint32_t wbase;
 
-  DEV void init(Bytes b, uint32_t p, uint32_t e) {
+  DEV void init(BS b, uint32_t p, uint32_t e) {
     B = b; pos = p; end = e;
     wbase = p;
-    __builtin_memcpy(&win, B.ptr(p), 8);
-    __builtin_memcpy(&win2, B.ptr(p) + 8, 8);
+    win = B.load8(p);
+    win2 = B.load8(p + 8);
   }
   // NOTE: refill may read up to 15 bytes past `end`; the host pads
   // every chunk with >= 8 newline bytes and 16B alignment slack
@@ -137,8 +177,8 @@ struct Cursor {
     uint32_t d = p - wbase;
     if (d >= 16u) {
       wbase = p;
-      __builtin_memcpy(&win, B.ptr(p), 8);
-      __builtin_memcpy(&win2, B.ptr(p) + 8, 8);
+      win = B.load8(p);
+      win2 = B.load8(p + 8);
       d = 0;
     }
     uint64_t w = d < 8u ? win : win2;
@@ -156,6 +196,7 @@ struct Cursor {
     }
   }
 };
+using Cursor = CursorT<Bytes>;
 
 // high bit set per zero byte; the FIRST flagged byte is always a true
 // zero (false positives only occur above a real zero byte)
@@ -173,21 +214,20 @@ DEV uint64_t str_special_mask(uint64_t w) {
 
 // Scan a JSON string body (cursor after the opening quote): SWAR over
 // 8-byte windows, per-byte only at escapes.  Sets raw off/len.
-DEV bool scan_string_fast(Cursor& c, uint32_t& off_out,
+template <class BS>
+DEV bool scan_string_fast(CursorT<BS>& c, uint32_t& off_out,
                           uint32_t& len_out) {
-  const uint8_t* d = c.B.mem - c.B.bias;
   uint32_t off = c.pos, p = c.pos, end = c.end;
   while (true) {
     while (p + 8 <= end) {
-      uint64_t w;
-      __builtin_memcpy(&w, d + p, 8);
+      uint64_t w = c.B.load8(p);
       uint64_t m = str_special_mask(w);
       if (m == 0) { p += 8; continue; }
       p += ((uint32_t)__ffsll((unsigned long long)m) - 1) >> 3;
       break;
     }
     if (p >= end) return false;
-    uint8_t b = d[p];
+    uint8_t b = c.B.at(p);
     if (b == '"') {
       off_out = off; len_out = p - off; c.pos = p + 1;
       return true;
@@ -195,11 +235,11 @@ DEV bool scan_string_fast(Cursor& c, uint32_t& off_out,
     if (b == '\\') {
       p++;
       if (p >= end) return false;
-      uint8_t e = d[p++];
+      uint8_t e = c.B.at(p); p++;
       if (e == 'u') {
         if (p + 4 > end) return false;
         for (int k = 0; k < 4; k++) {
-          uint8_t x = d[p + k];
+          uint8_t x = c.B.at(p + k);
           bool hex = (x >= '0' && x <= '9') || (x >= 'a' && x <= 'f') ||
                      (x >= 'A' && x <= 'F');
           if (!hex) return false;
@@ -219,13 +259,13 @@ DEV bool scan_string_fast(Cursor& c, uint32_t& off_out,
 // Component hash over zero-padded 8-byte words (mirrors
 // plan.comp_hash; any overread is masked away and stays inside the
 // padded buffer).
-DEV uint64_t comp_hash_span(Bytes B, uint32_t s, uint32_t e) {
+template <class BS>
+DEV uint64_t comp_hash_span(BS B, uint32_t s, uint32_t e) {
   uint64_t h = 0x9E3779B97F4A7C15ull;
   uint32_t len = e - s;
   uint32_t k = 0;
   do {
-    uint64_t w;
-    __builtin_memcpy(&w, B.ptr(s + k), 8);
+    uint64_t w = B.load8(s + k);
     uint32_t rem = len - k;
     if (rem < 8)
       w = (rem == 0) ? 0ull : (w & ((~0ull) >> (8 * (8 - rem))));
@@ -248,16 +288,15 @@ DEV uint64_t comp_hash_span(Bytes B, uint32_t s, uint32_t e) {
 // and krill pluck / synthetic sources never see.  Keys nested UNDER a
 // literal dotted key chain from the literal sig and match nothing.
 // Escaped keys are validated but get a sentinel signature.
-DEV int scan_key_sig(Cursor& c, uint64_t parent, uint64_t& sig_out,
+template <class BS>
+DEV int scan_key_sig(CursorT<BS>& c, uint64_t parent, uint64_t& sig_out,
                      bool dot_splits) {
-  const uint8_t* d = c.B.mem - c.B.bias;
   uint32_t p = c.pos, end = c.end;
   uint32_t comp_s = p;
   uint64_t sig = parent;
   while (true) {
     while (p + 8 <= end) {
-      uint64_t w;
-      __builtin_memcpy(&w, d + p, 8);
+      uint64_t w = c.B.load8(p);
       uint64_t m = str_special_mask(w) |
                    hz8(w ^ 0x2E2E2E2E2E2E2E2Eull);
       if (m == 0) { p += 8; continue; }
@@ -265,7 +304,7 @@ DEV int scan_key_sig(Cursor& c, uint64_t parent, uint64_t& sig_out,
       break;
     }
     if (p >= end) return 0;
-    uint8_t b = d[p];
+    uint8_t b = c.B.at(p);
     if (b == '"') {
       sig_out = mix64(sig ^ comp_hash_span(c.B, comp_s, p));
       c.pos = p + 1;
@@ -327,7 +366,8 @@ DEV double scale10(double v, long ex) {
 
 struct NumOut { double v; bool ok; };
 
-DEV NumOut parse_json_number(Cursor& c) {
+template <class BS>
+DEV NumOut parse_json_number(CursorT<BS>& c) {
   NumOut out; out.ok = false; out.v = 0.0;
   bool neg = false;
   if (!c.eof() && c.peek() == '-') { neg = true; c.pos++; }
@@ -390,32 +430,32 @@ DEV NumOut parse_json_number(Cursor& c) {
 
 // JavaScript ToNumber for record strings (mirrors krill.to_number):
 // trim ws; "" -> 0; decimal/hex/Infinity; else NaN.
-DEV double js_to_number(Bytes BV, uint32_t off, uint32_t len) {
-  const uint8_t* data = BV.mem - BV.bias;
+template <class BS>
+DEV double js_to_number(BS BV, uint32_t off, uint32_t len) {
   uint32_t i = 0, j = len;
-  while (i < j) { uint8_t b = data[off + i]; if (b==' '||b=='\t'||b=='\r'||b=='\n'||b=='\f'||b=='\v') i++; else break; }
-  while (j > i) { uint8_t b = data[off + j - 1]; if (b==' '||b=='\t'||b=='\r'||b=='\n'||b=='\f'||b=='\v') j--; else break; }
+  while (i < j) { uint8_t b = BV.at(off + i); if (b==' '||b=='\t'||b=='\r'||b=='\n'||b=='\f'||b=='\v') i++; else break; }
+  while (j > i) { uint8_t b = BV.at(off + j - 1); if (b==' '||b=='\t'||b=='\r'||b=='\n'||b=='\f'||b=='\v') j--; else break; }
   if (i == j) return 0.0;
   const double NAN_ = __builtin_nan("");
   uint32_t p = i;
   bool neg = false;
-  if (data[off+p] == '+' || data[off+p] == '-') { neg = data[off+p] == '-'; p++; }
+  if (BV.at(off+p) == '+' || BV.at(off+p) == '-') { neg = BV.at(off+p) == '-'; p++; }
   if (p == j) return NAN_;
   // Infinity
-  if (data[off+p] == 'I') {
+  if (BV.at(off+p) == 'I') {
     const char* inf = "Infinity";
     if (j - p == 8) {
-      for (int k = 0; k < 8; k++) if (data[off+p+k] != (uint8_t)inf[k]) return NAN_;
+      for (int k = 0; k < 8; k++) if (BV.at(off+p+k) != (uint8_t)inf[k]) return NAN_;
       return neg ? -__builtin_inf() : __builtin_inf();
     }
     return NAN_;
   }
   // hex
-  if (j - p > 2 && data[off+p] == '0' &&
-      (data[off+p+1] == 'x' || data[off+p+1] == 'X')) {
+  if (j - p > 2 && BV.at(off+p) == '0' &&
+      (BV.at(off+p+1) == 'x' || BV.at(off+p+1) == 'X')) {
     uint64_t v = 0;
     for (uint32_t k = p + 2; k < j; k++) {
-      uint8_t b = data[off+k];
+      uint8_t b = BV.at(off+k);
       uint32_t d;
       if (b >= '0' && b <= '9') d = b - '0';
       else if (b >= 'a' && b <= 'f') d = b - 'a' + 10;
@@ -428,29 +468,29 @@ DEV double js_to_number(Bytes BV, uint32_t off, uint32_t len) {
   }
   // decimal (JS grammar: digits [. digits] [e[+-]digits], '.5' and '5.' OK)
   uint64_t mant = 0; int ndig = 0, extra = 0; bool any = false;
-  while (p < j && data[off+p] >= '0' && data[off+p] <= '9') {
-    if (ndig < 19) { mant = mant * 10 + (data[off+p]-'0'); ndig++; }
+  while (p < j && BV.at(off+p) >= '0' && BV.at(off+p) <= '9') {
+    if (ndig < 19) { mant = mant * 10 + (BV.at(off+p)-'0'); ndig++; }
     else extra++;
     p++; any = true;
   }
-  if (p < j && data[off+p] == '.') {
+  if (p < j && BV.at(off+p) == '.') {
     p++;
-    while (p < j && data[off+p] >= '0' && data[off+p] <= '9') {
-      if (ndig < 19) { mant = mant * 10 + (data[off+p]-'0'); ndig++; extra--; }
+    while (p < j && BV.at(off+p) >= '0' && BV.at(off+p) <= '9') {
+      if (ndig < 19) { mant = mant * 10 + (BV.at(off+p)-'0'); ndig++; extra--; }
       p++; any = true;
     }
   }
   if (!any) return NAN_;
   long e10 = 0; int es = 1;
-  if (p < j && (data[off+p] == 'e' || data[off+p] == 'E')) {
+  if (p < j && (BV.at(off+p) == 'e' || BV.at(off+p) == 'E')) {
     p++;
-    if (p < j && (data[off+p] == '+' || data[off+p] == '-')) {
-      if (data[off+p] == '-') es = -1;
+    if (p < j && (BV.at(off+p) == '+' || BV.at(off+p) == '-')) {
+      if (BV.at(off+p) == '-') es = -1;
       p++;
     }
     if (p >= j) return NAN_;
-    while (p < j && data[off+p] >= '0' && data[off+p] <= '9') {
-      if (e10 < 100000) e10 = e10 * 10 + (data[off+p]-'0');
+    while (p < j && BV.at(off+p) >= '0' && BV.at(off+p) <= '9') {
+      if (e10 < 100000) e10 = e10 * 10 + (BV.at(off+p)-'0');
       p++;
     }
   }
@@ -478,19 +518,19 @@ DEV bool is_leap(long y) {
   return (y % 4 == 0) && ((y % 100 != 0) || (y % 400 == 0));
 }
 
-DEV DateOut parse_iso_ms(Bytes BV, uint32_t off, uint32_t len) {
-  const uint8_t* data = BV.mem - BV.bias;
+template <class BS>
+DEV DateOut parse_iso_ms(BS BV, uint32_t off, uint32_t len) {
   DateOut out; out.ok = false; out.ms = 0;
   // trim
   uint32_t i = 0, j = len;
-  while (i < j && (data[off+i]==' '||data[off+i]=='\t'||data[off+i]=='\r'||data[off+i]=='\n')) i++;
-  while (j > i && (data[off+j-1]==' '||data[off+j-1]=='\t'||data[off+j-1]=='\r'||data[off+j-1]=='\n')) j--;
+  while (i < j && (BV.at(off+i)==' '||BV.at(off+i)=='\t'||BV.at(off+i)=='\r'||BV.at(off+i)=='\n')) i++;
+  while (j > i && (BV.at(off+j-1)==' '||BV.at(off+j-1)=='\t'||BV.at(off+j-1)=='\r'||BV.at(off+j-1)=='\n')) j--;
   uint32_t p = i;
   auto digits = [&](int n, long& v) -> bool {
     v = 0;
     for (int k = 0; k < n; k++) {
       if (p >= j) return false;
-      uint8_t b = data[off+p];
+      uint8_t b = BV.at(off+p);
       if (b < '0' || b > '9') return false;
       v = v * 10 + (b - '0');
       p++;
@@ -500,27 +540,27 @@ DEV DateOut parse_iso_ms(Bytes BV, uint32_t off, uint32_t len) {
   long year, month = 1, day = 1, hh = 0, mm = 0, ss = 0, ms = 0;
   if (!digits(4, year)) return out;
   bool have_time = false;
-  if (p < j && data[off+p] == '-') {
+  if (p < j && BV.at(off+p) == '-') {
     p++;
     if (!digits(2, month)) return out;
-    if (p < j && data[off+p] == '-') {
+    if (p < j && BV.at(off+p) == '-') {
       p++;
       if (!digits(2, day)) return out;
-      if (p < j && (data[off+p] == 'T' || data[off+p] == ' ')) {
+      if (p < j && (BV.at(off+p) == 'T' || BV.at(off+p) == ' ')) {
         p++;
         if (!digits(2, hh)) return out;
-        if (p >= j || data[off+p] != ':') return out;
+        if (p >= j || BV.at(off+p) != ':') return out;
         p++;
         if (!digits(2, mm)) return out;
         have_time = true;
-        if (p < j && data[off+p] == ':') {
+        if (p < j && BV.at(off+p) == ':') {
           p++;
           if (!digits(2, ss)) return out;
-          if (p < j && data[off+p] == '.') {
+          if (p < j && BV.at(off+p) == '.') {
             p++;
             int nd = 0; long frac = 0;
-            while (p < j && data[off+p] >= '0' && data[off+p] <= '9' && nd < 9) {
-              if (nd < 3) frac = frac * 10 + (data[off+p] - '0');
+            while (p < j && BV.at(off+p) >= '0' && BV.at(off+p) <= '9' && nd < 9) {
+              if (nd < 3) frac = frac * 10 + (BV.at(off+p) - '0');
               nd++; p++;
             }
             if (nd == 0) return out;
@@ -533,14 +573,14 @@ DEV DateOut parse_iso_ms(Bytes BV, uint32_t off, uint32_t len) {
   }
   long tz_off_min = 0;
   if (have_time && p < j) {
-    uint8_t b = data[off+p];
+    uint8_t b = BV.at(off+p);
     if (b == 'Z') { p++; }
     else if (b == '+' || b == '-') {
       int sign = (b == '+') ? 1 : -1;
       p++;
       long th, tm;
       if (!digits(2, th)) return out;
-      if (p < j && data[off+p] == ':') p++;
+      if (p < j && BV.at(off+p) == ':') p++;
       if (!digits(2, tm)) return out;
       tz_off_min = sign * (th * 60 + tm);
     }
@@ -564,10 +604,11 @@ DEV DateOut parse_iso_ms(Bytes BV, uint32_t off, uint32_t len) {
 // Parses one record (bytes [start,end)); fills fv (all slots must be
 // preinitialized to T_MISSING by the caller).  Returns false on invalid
 // JSON.  top_type receives the top-level value type.
-DEV bool parse_record(Bytes BV, uint32_t start, uint32_t end,
+template <class BS>
+DEV bool parse_record(BS BV, uint32_t start, uint32_t end,
                       const PlanView& P, FV& fv, uint8_t& top_type,
                       uint64_t* sig_lds, bool dot_splits) {
-  Cursor c; c.init(BV, start, end);
+  CursorT<BS> c; c.init(BV, start, end);
   // parent-signature stack in LDS: [depth * BLOCK + tid]
   #define sig_stack_at(d) sig_lds[(d) * BLOCK + (uint32_t)threadIdx.x]
 
@@ -781,12 +822,13 @@ DEV int hexval4(uint8_t x) {
 // surrogate pairs included): packs 1-4 UTF-8 bytes little-endian into
 // out, advances i, returns the byte count (0 = bad escape — cannot
 // happen for spans the tokenizer accepted; defensive).
-DEV int unesc_next(const uint8_t* d, uint32_t fo, uint32_t fl,
+template <class BS>
+DEV int unesc_next(BS d, uint32_t fo, uint32_t fl,
                    uint32_t& i, uint32_t& out) {
-  uint8_t b = d[fo + i];
+  uint8_t b = d.at(fo + i);
   if (b != '\\') { i++; out = b; return 1; }
   if (i + 2 > fl) return 0;
-  uint8_t e = d[fo + i + 1];
+  uint8_t e = d.at(fo + i + 1);
   i += 2;
   switch (e) {
     case '"':  out = '"';  return 1;
@@ -803,17 +845,17 @@ DEV int unesc_next(const uint8_t* d, uint32_t fo, uint32_t fl,
   if (i + 4 > fl) return 0;
   uint32_t cp = 0;
   for (int k = 0; k < 4; k++) {
-    int h = hexval4(d[fo + i + k]);
+    int h = hexval4(d.at(fo + i + k));
     if (h < 0) return 0;
     cp = cp * 16 + (uint32_t)h;
   }
   i += 4;
   if (cp >= 0xD800 && cp < 0xDC00 && i + 6 <= fl &&
-      d[fo + i] == '\\' && d[fo + i + 1] == 'u') {
+      d.at(fo + i) == '\\' && d.at(fo + i + 1) == 'u') {
     uint32_t lo = 0;
     bool ok = true;
     for (int k = 0; k < 4; k++) {
-      int h = hexval4(d[fo + i + 2 + k]);
+      int h = hexval4(d.at(fo + i + 2 + k));
       if (h < 0) { ok = false; break; }
       lo = lo * 16 + (uint32_t)h;
     }
@@ -840,7 +882,8 @@ DEV int unesc_next(const uint8_t* d, uint32_t fo, uint32_t fl,
 
 // 3-way compare of the DECODED field span vs const bytes; -2 on bad
 // escape (defensive — the tokenizer already rejected those records).
-DEV int unesc_cmp(const uint8_t* d, uint32_t fo, uint32_t fl,
+template <class BS>
+DEV int unesc_cmp(BS d, uint32_t fo, uint32_t fl,
                   const uint8_t* cb, uint32_t cl) {
   uint32_t i = 0, j = 0;
   while (i < fl) {
@@ -857,18 +900,19 @@ DEV int unesc_cmp(const uint8_t* d, uint32_t fo, uint32_t fl,
   return (j == cl) ? 0 : -1;
 }
 
-DEV bool span_has_backslash(const uint8_t* d, uint32_t fo, uint32_t fl) {
+template <class BS>
+DEV bool span_has_backslash(BS d, uint32_t fo, uint32_t fl) {
   for (uint32_t k = 0; k < fl; k++)
-    if (d[fo + k] == '\\') return true;
+    if (d.at(fo + k) == '\\') return true;
   return false;
 }
 
 // -------------------------------------------------------------------
 // predicate evaluation (K2): -1 throw (missing field), 0 false, 1 true
 
-DEV int eval_leaf(const PlanView& P, Bytes BV, const FV& fv,
+template <class BS>
+DEV int eval_leaf(const PlanView& P, BS BV, const FV& fv,
                   int op, int slot, int cidx) {
-  const uint8_t* data = BV.mem - BV.bias;
   uint8_t ft = fv.get_type(slot);
   if (ft == T_MISSING) return -1;  // krill: missing field -> throw
 
@@ -894,7 +938,7 @@ DEV int eval_leaf(const PlanView& P, Bytes BV, const FV& fv,
         if (fl == clen) {
           eq = true;
           for (uint32_t k = 0; k < fl; k++)
-            if (data[fo + k] != P.const_bytes[coff + k]) { eq = false; break; }
+            if (BV.at(fo + k) != P.const_bytes[coff + k]) { eq = false; break; }
         }
         // records may carry the constant in JSON-ESCAPED form: try
         // the canonical escaped rendering too (plan.py ConstPool)
@@ -904,13 +948,13 @@ DEV int eval_leaf(const PlanView& P, Bytes BV, const FV& fv,
           if (cl2 != 0 && fl == cl2) {
             eq = true;
             for (uint32_t k = 0; k < fl; k++)
-              if (data[fo + k] != P.const_bytes[co2 + k]) { eq = false; break; }
+              if (BV.at(fo + k) != P.const_bytes[co2 + k]) { eq = false; break; }
           }
         }
         // NON-canonical escapes in the data (GET, \/, surrogate
         // pairs): unescape-as-you-compare (cold; escaped spans only)
-        if (!eq && span_has_backslash(data, fo, fl))
-          eq = unesc_cmp(data, fo, fl, P.const_bytes + coff, clen) == 0;
+        if (!eq && span_has_backslash(BV, fo, fl))
+          eq = unesc_cmp(BV, fo, fl, P.const_bytes + coff, clen) == 0;
       } else {  // string vs number: ToNumber(field)
         double fn = js_to_number(BV, fv.get_soff(slot), fv.get_slen(slot));
         eq = (fn == fn) && (fn == cdval);
@@ -925,13 +969,13 @@ DEV int eval_leaf(const PlanView& P, Bytes BV, const FV& fv,
   if (ft == T_STR && ckind == CONST_STR) {
     uint32_t fo = fv.get_soff(slot), fl = fv.get_slen(slot);
     int cmp = 0;
-    if (span_has_backslash(data, fo, fl)) {
-      cmp = unesc_cmp(data, fo, fl, P.const_bytes + coff, clen);
+    if (span_has_backslash(BV, fo, fl)) {
+      cmp = unesc_cmp(BV, fo, fl, P.const_bytes + coff, clen);
       if (cmp == -2) cmp = 0;  // unreachable: tokenizer validated
     } else {
       uint32_t n = fl < clen ? fl : clen;
       for (uint32_t k = 0; k < n; k++) {
-        uint8_t a = data[fo + k], b = P.const_bytes[coff + k];
+        uint8_t a = BV.at(fo + k), b = P.const_bytes[coff + k];
         if (a != b) { cmp = a < b ? -1 : 1; break; }
       }
       if (cmp == 0) cmp = (fl < clen) ? -1 : (fl > clen ? 1 : 0);
@@ -962,7 +1006,8 @@ DEV int eval_leaf(const PlanView& P, Bytes BV, const FV& fv,
   }
 }
 
-DEV int eval_predicate(const PlanView& P, Bytes BV,
+template <class BS>
+DEV int eval_predicate(const PlanView& P, BS BV,
                        const FV& fv, int prog_id) {
   int idx = P.prog_bounds[prog_id * 2 + 0];
   // frame packed into u64: op(1b) | remaining(23b) | end(32b), kept in
@@ -1014,17 +1059,17 @@ DEV int eval_predicate(const PlanView& P, Bytes BV,
 // -------------------------------------------------------------------
 // dictionaries (string + number interning)
 
-DEV uint64_t hash_bytes(Bytes BV, uint32_t off, uint32_t len) {
-  const uint8_t* data = BV.mem - BV.bias;
+template <class BS>
+DEV uint64_t hash_bytes(BS BV, uint32_t off, uint32_t len) {
   uint64_t h = FNV_OFFSET;
-  for (uint32_t k = 0; k < len; k++) h = fnv1a_byte(h, data[off + k]);
+  for (uint32_t k = 0; k < len; k++) h = fnv1a_byte(h, BV.at(off + k));
   return mix64(h ^ len);
 }
 
 // Returns string id, or 0xFFFFFFFF on table/data overflow.
-DEV uint32_t intern_string(const StrDict& D, Bytes BV,
+template <class BS>
+DEV uint32_t intern_string(const StrDict& D, BS BV,
                            uint32_t off, uint32_t len) {
-  const uint8_t* data = BV.mem - BV.bias;
   uint64_t h = hash_bytes(BV, off, len);
   uint32_t mask = D.nslots - 1;
   uint32_t s = (uint32_t)h & mask;
@@ -1039,7 +1084,7 @@ DEV uint32_t intern_string(const StrDict& D, Bytes BV,
         if (l2 != len) break;
         bool same = true;
         for (uint32_t k = 0; k < len; k++)
-          if (D.data[o2 + k] != data[off + k]) { same = false; break; }
+          if (D.data[o2 + k] != BV.at(off + k)) { same = false; break; }
         if (same) return atomic_load_relaxed(&D.id[s]);
         break;
       }
@@ -1050,7 +1095,7 @@ DEV uint32_t intern_string(const StrDict& D, Bytes BV,
           uint32_t o = atomicAdd(D.data_used, (len + 7u) & ~7u);
           if (o + len > D.data_cap) return 0xFFFFFFFFu;  // overflow
           for (uint32_t k = 0; k < len; k++)
-            atomic_store_relaxed(&D.data[o + k], data[off + k]);
+            atomic_store_relaxed(&D.data[o + k], BV.at(off + k));
           uint32_t myid = atomicAdd(D.next_id, 1u);
           atomic_store_relaxed(&D.hash[s], h);
           atomic_store_relaxed(&D.off[s], o);
@@ -1261,12 +1306,13 @@ __global__ void newline_write_kernel(const uint8_t* data,
   }
 }
 
+template <int XP>
 DEV void scan_kernel_body(char* smem, ScanArgs A);
 
 __launch_bounds__(BLOCK)
 __global__ void scan_kernel(ScanArgs A) {
   extern __shared__ __attribute__((aligned(16))) char smem0[];
-  scan_kernel_body(smem0, A);
+  scan_kernel_body<0>(smem0, A);
 }
 
 // occupancy-capped variants (min waves per SIMD; constrains the
@@ -1275,7 +1321,7 @@ template <int MW>
 __launch_bounds__(BLOCK, MW)
 __global__ void scan_kernel_mw(ScanArgs A) {
   extern __shared__ __attribute__((aligned(16))) char smem1[];
-  scan_kernel_body(smem1, A);
+  scan_kernel_body<0>(smem1, A);
 }
 template __global__ void scan_kernel_mw<2>(ScanArgs);
 template __global__ void scan_kernel_mw<3>(ScanArgs);
@@ -1283,6 +1329,268 @@ template __global__ void scan_kernel_mw<4>(ScanArgs);
 template __global__ void scan_kernel_mw<5>(ScanArgs);
 template __global__ void scan_kernel_mw<6>(ScanArgs);
 
+// wave-transposed record staging (XBytes; DRAGNET_XPOSE)
+template <int MW>
+__launch_bounds__(BLOCK, MW)
+__global__ void scan_kernel_x(ScanArgs A) {
+  extern __shared__ __attribute__((aligned(16))) char smem2[];
+  scan_kernel_body<1>(smem2, A);
+}
+template __global__ void scan_kernel_x<2>(ScanArgs);
+template __global__ void scan_kernel_x<3>(ScanArgs);
+template __global__ void scan_kernel_x<4>(ScanArgs);
+
+// Per-record pipeline (K1-K6) shared by the linear and the
+// wave-transposed (XBytes) kernels: tokenize/extract, predicate,
+// synthetics, per-metric filter + bucketize + intern + aggregate.
+#define synth_val_at(si) synth_lds[(si) * BLOCK + threadIdx.x]
+
+template <class BS>
+DEV void process_record(BS BV, uint32_t start, uint32_t end,
+                        const ScanArgs& A, const PlanView& P, FV& fv,
+                        unsigned long long* lcnt, LdsCacheEntry* cache,
+                        uint64_t* sig_lds, double* synth_lds) {
+  const int nf = P.nf;
+  uint32_t synth_ok = 0;
+  atomicAdd(&lcnt[C_LINES], 1ull);
+  // only the type lane needs initializing: soff/slen/num are read
+  // only after a capture set them
+  for (int f = 0; f < nf; f++)
+    fv.type[f * BLOCK + fv.tid] = T_MISSING;
+
+  uint8_t top_type;
+  bool ok = (end > start) &&
+            parse_record(BV, start, end, P, fv, top_type, sig_lds,
+                         A.data_format_skinner);
+  double weight = 1.0;
+  if (ok && A.data_format_skinner) {
+    // require: object top, a "fields" member, numeric "value"
+    bool has_fields = P.fields_slot >= 0 &&
+                      fv.get_type(P.fields_slot) != T_MISSING;
+    bool val_num = P.value_slot >= 0 &&
+                   fv.get_type(P.value_slot) == T_NUM;
+    if (top_type != T_OBJ || !has_fields || !val_num) ok = false;
+    else weight = fv.get_num(P.value_slot);
+  }
+  if (!ok) {
+    atomicAdd(&lcnt[C_INVALID_JSON], 1ull);
+  } else {
+    atomicAdd(&lcnt[C_PARSED], 1ull);
+
+    // datasource filter (program 0)
+    int keep = eval_predicate(P, BV, fv, 0);
+    if (keep == -1) atomicAdd(&lcnt[C_DS_FAILEDEVAL], 1ull);
+    else if (keep == 0) atomicAdd(&lcnt[C_DS_FILTERED], 1ull);
+
+    if (keep == 1) {
+      // synthetic date fields (shared across metrics)
+      synth_ok = 0;
+      for (int si = 0; si < P.ns; si++) {
+        int slot = P.synth_slots[si];
+        uint8_t t = fv.get_type(slot);
+        uint32_t ok;
+        if (t == T_MISSING) ok = 2;                           // undef
+        else if (t == T_NUM) {
+          ok = 1; synth_val_at(si) = fv.get_num(slot);
+        } else if (t == T_STR) {
+          DateOut d = parse_iso_ms(BV, fv.get_soff(slot),
+                                   fv.get_slen(slot));
+          if (d.ok) {
+            long long secs = d.ms >= 0 ? d.ms / 1000
+                                       : (d.ms - 999) / 1000;
+            ok = 1; synth_val_at(si) = (double)secs;
+          } else ok = 3;                                      // baddate
+        } else ok = 3;  // bool/null/obj/arr: Date.parse fails
+        synth_ok |= ok << (2 * si);
+      }
+
+      // per-metric pipeline
+      for (int m = 0; m < P.nm; m++) {
+        const int32_t* M = &P.metric_rows[m * 8];
+        unsigned long long* mc = &lcnt[C_GLOBAL_N + m * CM_N];
+        atomicAdd(&mc[CM_FILTER_IN], 1ull);
+
+        int res = eval_predicate(P, BV, fv, M[0]);
+        if (res == -1) { atomicAdd(&mc[CM_FAILEDEVAL], 1ull); continue; }
+        if (res == 0) { atomicAdd(&mc[CM_FILTERED], 1ull); continue; }
+
+        // synthetic requirements (first failure counted; record
+        // dropped on any failure — stream-synthetic.js:37-85)
+        bool sok = true;
+        for (int k = 0; k < M[3]; k++) {
+          int si = P.synth_req[M[4] + k];
+          uint32_t ok = (synth_ok >> (2 * si)) & 3u;
+          if (ok != 1) {
+            atomicAdd(&mc[ok == 2 ? CM_UNDEF : CM_BADDATE], 1ull);
+            sok = false; break;
+          }
+        }
+        if (!sok) continue;
+
+        // time filter on dn_ts (= last synth req when present)
+        if (M[5]) {
+          int si = P.synth_req[M[4] + M[3] - 1];
+          double ts = synth_val_at(si);
+          if (!(ts >= (double)M[6] && ts < (double)M[7])) {
+            atomicAdd(&mc[CM_TIME_OUT], 1ull); continue;
+          }
+        }
+
+        atomicAdd(&mc[CM_AGG_IN], 1ull);
+
+        // build the group key (K4: bucketize; dict-intern)
+        uint32_t key[MAX_KEY];
+        int nk = M[1];
+        bool drop = false, overflow = false;
+        for (int bi = 0; bi < nk; bi++) {
+          const int32_t* B = &P.bd_rows[(M[2] + bi) * 4];
+          double step = P.bd_steps[M[2] + bi];
+          uint8_t t; double num = 0.0; uint32_t so = 0, sl = 0;
+          if (B[0] == 1) {  // synthetic date value
+            int si = B[1];
+            if (((synth_ok >> (2 * si)) & 3u) == 1) {
+              t = T_NUM; num = synth_val_at(si);
+            } else t = T_MISSING;  // cannot happen: required above
+          } else {
+            int slot = B[1];
+            // aggregation lookup is literal-first (points.lookup):
+            // a top-level literal dotted key (companion slot)
+            // shadows the plucked nested value
+            int cs2 = P.comp_slot[slot];
+            if (cs2 >= 0 && fv.get_type(cs2) != T_MISSING)
+              slot = cs2;
+            t = fv.get_type(slot);
+            num = fv.get_num(slot);
+            so = fv.get_soff(slot); sl = fv.get_slen(slot);
+          }
+          uint32_t code;
+          if (B[2] != BUCKET_NONE) {
+            if (t != T_NUM) { drop = true; break; }  // nonnumeric
+            long long ord;
+            if (B[2] == BUCKET_P2) {
+              if (!(num >= 1.0)) ord = 0;
+              else {
+                // floor(log2(v)) + 1 via exponent field
+                uint64_t bits = __double_as_longlong(num);
+                ord = (long long)((bits >> 52) & 0x7FF) - 1023 + 1;
+              }
+            } else {
+              ord = (long long)__builtin_floor(num / step);
+            }
+            if (ord >= -(long long)ORD_BIAS && ord < (long long)ORD_BIAS) {
+              code = make_code(TAG_ORD, (uint32_t)(ord + ORD_BIAS));
+            } else {
+              uint32_t id = intern_number(A.ndict, (double)ord);
+              if (id == 0xFFFFFFFFu) { overflow = true; break; }
+              code = make_code(TAG_NUM, id);
+            }
+          } else if (t == T_MISSING) {
+            code = make_code(TAG_SPECIAL, SPECIAL_UNDEF);
+          } else if (t == T_NULL) {
+            code = make_code(TAG_SPECIAL, SPECIAL_NULL);
+          } else if (t == T_TRUE) {
+            code = make_code(TAG_SPECIAL, SPECIAL_TRUE);
+          } else if (t == T_FALSE) {
+            code = make_code(TAG_SPECIAL, SPECIAL_FALSE);
+          } else if (t == T_OBJ) {
+            code = make_code(TAG_SPECIAL, SPECIAL_OBJECT);
+          } else if (t == T_ARR) {
+            // intern the raw JSON span; the host canonicalizes it
+            // with JS Array.toString semantics (plan.decode_key)
+            uint32_t id = intern_string(A.sdict, BV, so, sl);
+            if (id == 0xFFFFFFFFu || id >= (1u << 27)) {
+              overflow = true; break;
+            }
+            code = make_code(TAG_SPECIAL, SPECIAL_ARRJSON | (id << 3));
+          } else if (t == T_NUM) {
+            uint32_t id = intern_number(A.ndict, num);
+            if (id == 0xFFFFFFFFu) { overflow = true; break; }
+            code = make_code(TAG_NUM, id);
+          } else {  // T_STR
+            uint32_t id = intern_string(A.sdict, BV, so, sl);
+            if (id == 0xFFFFFFFFu) { overflow = true; break; }
+            code = make_code(TAG_STR, id);
+          }
+          // constant-index write keeps key[] in registers
+#pragma unroll
+          for (int kk = 0; kk < MAX_KEY; kk++)
+            if (kk == bi) key[kk] = code;
+        }
+        if (overflow) { atomicAdd(&lcnt[C_OVERFLOW], 1ull); continue; }
+        if (drop) { atomicAdd(&mc[CM_NONNUMERIC], 1ull); continue; }
+#pragma unroll
+        for (int k = 0; k < MAX_KEY; k++)
+          if (k >= nk) key[k] = 0;
+
+        // LDS combining cache: hash (metric, key)
+        uint64_t kh = mix64((uint64_t)m * 0x9E3779B97F4A7C15ull + 1);
+        for (int k = 0; k < MAX_KEY; k++) kh = mix64(kh ^ key[k]);
+        if (kh == 0) kh = 1;
+
+        // Wave-level pre-combining: lanes carrying the same key
+        // elect one leader per distinct kh which adds the whole
+        // group's weight — one LDS atomic per distinct key per
+        // wavefront instead of one per record.  (Only when every
+        // weight is 1: json format; skinner weights vary.)
+        if (!A.data_format_skinner) {
+          uint64_t unproc = __ballot(true);  // lanes still here
+          bool leader = false;
+          double wsum = 0.0;
+          while (unproc) {
+            int src = (int)(__ffsll((long long)unproc) - 1);
+            uint64_t src_kh = __shfl(kh, src);
+            bool same = (kh == src_kh);
+            uint64_t grp = __ballot(same);
+            if (same && __lane_id() == src) {
+              leader = true;
+              wsum = (double)__popcll(grp);
+            }
+            unproc &= ~grp;
+          }
+          if (!leader) continue;  // combined into the leader's add
+          weight = wsum;
+        }
+        bool cached = false;
+        uint32_t ci = (uint32_t)kh & (LDS_CACHE - 1);
+        // 8 probes: a hot key that loses every probe degrades to
+        // per-record contended global inserts (measured 4x slower
+        // on low-cardinality ordinal breakdowns with 2 probes)
+        for (int attempt = 0; attempt < 8; attempt++) {
+          unsigned long long prev = atomicCAS(
+              (unsigned long long*)&cache[ci].hash, 0ull,
+              (unsigned long long)kh);
+          if (prev == 0) {
+            // claimed: fill identity (hash claim is the sync point;
+            // same-key lanes re-check identity below)
+            cache[ci].metric = m;
+            for (int k = 0; k < MAX_KEY; k++) cache[ci].key[k] = key[k];
+            atomicAdd(&cache[ci].count, weight);
+            cached = true;
+            break;
+          }
+          if (prev == (unsigned long long)kh &&
+              cache[ci].metric == m) {
+            bool same = true;
+            for (int k = 0; k < MAX_KEY; k++)
+              if (cache[ci].key[k] != key[k]) { same = false; break; }
+            if (same) {
+              atomicAdd(&cache[ci].count, weight);
+              cached = true;
+              break;
+            }
+          }
+          ci = (ci + 1) & (LDS_CACHE - 1);
+        }
+        if (!cached) {
+          if (!agg_insert(A.tables[m], key, nk, weight))
+            atomicAdd(&lcnt[C_OVERFLOW], 1ull);
+        }
+      }
+    }
+  }
+}
+
+template <int XP>
 DEV void scan_kernel_body(char* smem, ScanArgs A) {
   const PlanView& P = A.P;
   const int nf = P.nf;
@@ -1324,10 +1632,28 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
   fv.type = fv_type; fv.soff = fv_soff; fv.slen = fv_slen; fv.num = fv_num;
   fv.tid = threadIdx.x;
 
-  // synthetic values live in LDS (keeps the MW=4 build spill-free)
-  #define synth_val_at(si) synth_lds[(si) * BLOCK + threadIdx.x]
-  uint32_t synth_ok = 0;  // 2 bits per synthetic field
-
+  if constexpr (XP == 1) {
+    // wave-transposed pool: slot r's bytes live granule-interleaved
+    // at xwave_base[r/64] + (r%64)*XGRAN; a wave's refills are 64
+    // consecutive granules (coalesced) instead of 64 scattered
+    // records.  Record order is length-sorted by the host — the
+    // aggregation is order-independent (associative merge).
+    const uint32_t stride_x = gridDim.x * BLOCK;
+    for (uint32_t rbase = blockIdx.x * BLOCK; rbase < A.xn_slots;
+         rbase += stride_x) {
+      uint32_t r = rbase + threadIdx.x;
+      if (r < A.xn_slots) {
+        uint32_t len = A.xrec_len[r];
+        if (len != 0xFFFFFFFFu) {
+          XBytes BV;
+          BV.base = A.xdata + A.xwave_base[r >> 6]
+                    + (size_t)(r & 63u) * XGRAN;
+          process_record(BV, 0u, len, A, P, fv, lcnt, cache, sig_lds,
+                         synth_lds);
+        }
+      }
+    }
+  } else {
   uint32_t nlines = *A.nlines_ptr;
   if (nlines > A.pos_cap) nlines = A.pos_cap;
   const uint32_t stride = gridDim.x * BLOCK;
@@ -1361,248 +1687,15 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
     }
 
     if (active) {
-      atomicAdd(&lcnt[C_LINES], 1ull);
-      // only the type lane needs initializing: soff/slen/num are read
-      // only after a capture set them
-      for (int f = 0; f < nf; f++)
-        fv.type[f * BLOCK + fv.tid] = T_MISSING;
-
       uint32_t start = r ? A.nl_pos[r - 1] + 1 : A.first_start;
       uint32_t end = A.nl_pos[r];
-      uint8_t top_type;
-      bool ok = (end > start) &&
-                parse_record(BV, start, end, P, fv, top_type, sig_lds,
-                             A.data_format_skinner);
-      double weight = 1.0;
-      if (ok && A.data_format_skinner) {
-        // require: object top, a "fields" member, numeric "value"
-        bool has_fields = P.fields_slot >= 0 &&
-                          fv.get_type(P.fields_slot) != T_MISSING;
-        bool val_num = P.value_slot >= 0 &&
-                       fv.get_type(P.value_slot) == T_NUM;
-        if (top_type != T_OBJ || !has_fields || !val_num) ok = false;
-        else weight = fv.get_num(P.value_slot);
-      }
-      if (!ok) {
-        atomicAdd(&lcnt[C_INVALID_JSON], 1ull);
-      } else {
-        atomicAdd(&lcnt[C_PARSED], 1ull);
-
-        // datasource filter (program 0)
-        int keep = eval_predicate(P, BV, fv, 0);
-        if (keep == -1) atomicAdd(&lcnt[C_DS_FAILEDEVAL], 1ull);
-        else if (keep == 0) atomicAdd(&lcnt[C_DS_FILTERED], 1ull);
-
-        if (keep == 1) {
-          // synthetic date fields (shared across metrics)
-          synth_ok = 0;
-          for (int si = 0; si < P.ns; si++) {
-            int slot = P.synth_slots[si];
-            uint8_t t = fv.get_type(slot);
-            uint32_t ok;
-            if (t == T_MISSING) ok = 2;                           // undef
-            else if (t == T_NUM) {
-              ok = 1; synth_val_at(si) = fv.get_num(slot);
-            } else if (t == T_STR) {
-              DateOut d = parse_iso_ms(BV, fv.get_soff(slot),
-                                       fv.get_slen(slot));
-              if (d.ok) {
-                long long secs = d.ms >= 0 ? d.ms / 1000
-                                           : (d.ms - 999) / 1000;
-                ok = 1; synth_val_at(si) = (double)secs;
-              } else ok = 3;                                      // baddate
-            } else ok = 3;  // bool/null/obj/arr: Date.parse fails
-            synth_ok |= ok << (2 * si);
-          }
-
-          // per-metric pipeline
-          for (int m = 0; m < P.nm; m++) {
-            const int32_t* M = &P.metric_rows[m * 8];
-            unsigned long long* mc = &lcnt[C_GLOBAL_N + m * CM_N];
-            atomicAdd(&mc[CM_FILTER_IN], 1ull);
-
-            int res = eval_predicate(P, BV, fv, M[0]);
-            if (res == -1) { atomicAdd(&mc[CM_FAILEDEVAL], 1ull); continue; }
-            if (res == 0) { atomicAdd(&mc[CM_FILTERED], 1ull); continue; }
-
-            // synthetic requirements (first failure counted; record
-            // dropped on any failure — stream-synthetic.js:37-85)
-            bool sok = true;
-            for (int k = 0; k < M[3]; k++) {
-              int si = P.synth_req[M[4] + k];
-              uint32_t ok = (synth_ok >> (2 * si)) & 3u;
-              if (ok != 1) {
-                atomicAdd(&mc[ok == 2 ? CM_UNDEF : CM_BADDATE], 1ull);
-                sok = false; break;
-              }
-            }
-            if (!sok) continue;
-
-            // time filter on dn_ts (= last synth req when present)
-            if (M[5]) {
-              int si = P.synth_req[M[4] + M[3] - 1];
-              double ts = synth_val_at(si);
-              if (!(ts >= (double)M[6] && ts < (double)M[7])) {
-                atomicAdd(&mc[CM_TIME_OUT], 1ull); continue;
-              }
-            }
-
-            atomicAdd(&mc[CM_AGG_IN], 1ull);
-
-            // build the group key (K4: bucketize; dict-intern)
-            uint32_t key[MAX_KEY];
-            int nk = M[1];
-            bool drop = false, overflow = false;
-            for (int bi = 0; bi < nk; bi++) {
-              const int32_t* B = &P.bd_rows[(M[2] + bi) * 4];
-              double step = P.bd_steps[M[2] + bi];
-              uint8_t t; double num = 0.0; uint32_t so = 0, sl = 0;
-              if (B[0] == 1) {  // synthetic date value
-                int si = B[1];
-                if (((synth_ok >> (2 * si)) & 3u) == 1) {
-                  t = T_NUM; num = synth_val_at(si);
-                } else t = T_MISSING;  // cannot happen: required above
-              } else {
-                int slot = B[1];
-                // aggregation lookup is literal-first (points.lookup):
-                // a top-level literal dotted key (companion slot)
-                // shadows the plucked nested value
-                int cs2 = P.comp_slot[slot];
-                if (cs2 >= 0 && fv.get_type(cs2) != T_MISSING)
-                  slot = cs2;
-                t = fv.get_type(slot);
-                num = fv.get_num(slot);
-                so = fv.get_soff(slot); sl = fv.get_slen(slot);
-              }
-              uint32_t code;
-              if (B[2] != BUCKET_NONE) {
-                if (t != T_NUM) { drop = true; break; }  // nonnumeric
-                long long ord;
-                if (B[2] == BUCKET_P2) {
-                  if (!(num >= 1.0)) ord = 0;
-                  else {
-                    // floor(log2(v)) + 1 via exponent field
-                    uint64_t bits = __double_as_longlong(num);
-                    ord = (long long)((bits >> 52) & 0x7FF) - 1023 + 1;
-                  }
-                } else {
-                  ord = (long long)__builtin_floor(num / step);
-                }
-                if (ord >= -(long long)ORD_BIAS && ord < (long long)ORD_BIAS) {
-                  code = make_code(TAG_ORD, (uint32_t)(ord + ORD_BIAS));
-                } else {
-                  uint32_t id = intern_number(A.ndict, (double)ord);
-                  if (id == 0xFFFFFFFFu) { overflow = true; break; }
-                  code = make_code(TAG_NUM, id);
-                }
-              } else if (t == T_MISSING) {
-                code = make_code(TAG_SPECIAL, SPECIAL_UNDEF);
-              } else if (t == T_NULL) {
-                code = make_code(TAG_SPECIAL, SPECIAL_NULL);
-              } else if (t == T_TRUE) {
-                code = make_code(TAG_SPECIAL, SPECIAL_TRUE);
-              } else if (t == T_FALSE) {
-                code = make_code(TAG_SPECIAL, SPECIAL_FALSE);
-              } else if (t == T_OBJ) {
-                code = make_code(TAG_SPECIAL, SPECIAL_OBJECT);
-              } else if (t == T_ARR) {
-                // intern the raw JSON span; the host canonicalizes it
-                // with JS Array.toString semantics (plan.decode_key)
-                uint32_t id = intern_string(A.sdict, BV, so, sl);
-                if (id == 0xFFFFFFFFu || id >= (1u << 27)) {
-                  overflow = true; break;
-                }
-                code = make_code(TAG_SPECIAL, SPECIAL_ARRJSON | (id << 3));
-              } else if (t == T_NUM) {
-                uint32_t id = intern_number(A.ndict, num);
-                if (id == 0xFFFFFFFFu) { overflow = true; break; }
-                code = make_code(TAG_NUM, id);
-              } else {  // T_STR
-                uint32_t id = intern_string(A.sdict, BV, so, sl);
-                if (id == 0xFFFFFFFFu) { overflow = true; break; }
-                code = make_code(TAG_STR, id);
-              }
-              // constant-index write keeps key[] in registers
-#pragma unroll
-              for (int kk = 0; kk < MAX_KEY; kk++)
-                if (kk == bi) key[kk] = code;
-            }
-            if (overflow) { atomicAdd(&lcnt[C_OVERFLOW], 1ull); continue; }
-            if (drop) { atomicAdd(&mc[CM_NONNUMERIC], 1ull); continue; }
-#pragma unroll
-            for (int k = 0; k < MAX_KEY; k++)
-              if (k >= nk) key[k] = 0;
-
-            // LDS combining cache: hash (metric, key)
-            uint64_t kh = mix64((uint64_t)m * 0x9E3779B97F4A7C15ull + 1);
-            for (int k = 0; k < MAX_KEY; k++) kh = mix64(kh ^ key[k]);
-            if (kh == 0) kh = 1;
-
-            // Wave-level pre-combining: lanes carrying the same key
-            // elect one leader per distinct kh which adds the whole
-            // group's weight — one LDS atomic per distinct key per
-            // wavefront instead of one per record.  (Only when every
-            // weight is 1: json format; skinner weights vary.)
-            if (!A.data_format_skinner) {
-              uint64_t unproc = __ballot(true);  // lanes still here
-              bool leader = false;
-              double wsum = 0.0;
-              while (unproc) {
-                int src = (int)(__ffsll((long long)unproc) - 1);
-                uint64_t src_kh = __shfl(kh, src);
-                bool same = (kh == src_kh);
-                uint64_t grp = __ballot(same);
-                if (same && __lane_id() == src) {
-                  leader = true;
-                  wsum = (double)__popcll(grp);
-                }
-                unproc &= ~grp;
-              }
-              if (!leader) continue;  // combined into the leader's add
-              weight = wsum;
-            }
-            bool cached = false;
-            uint32_t ci = (uint32_t)kh & (LDS_CACHE - 1);
-            // 8 probes: a hot key that loses every probe degrades to
-            // per-record contended global inserts (measured 4x slower
-            // on low-cardinality ordinal breakdowns with 2 probes)
-            for (int attempt = 0; attempt < 8; attempt++) {
-              unsigned long long prev = atomicCAS(
-                  (unsigned long long*)&cache[ci].hash, 0ull,
-                  (unsigned long long)kh);
-              if (prev == 0) {
-                // claimed: fill identity (hash claim is the sync point;
-                // same-key lanes re-check identity below)
-                cache[ci].metric = m;
-                for (int k = 0; k < MAX_KEY; k++) cache[ci].key[k] = key[k];
-                atomicAdd(&cache[ci].count, weight);
-                cached = true;
-                break;
-              }
-              if (prev == (unsigned long long)kh &&
-                  cache[ci].metric == m) {
-                bool same = true;
-                for (int k = 0; k < MAX_KEY; k++)
-                  if (cache[ci].key[k] != key[k]) { same = false; break; }
-                if (same) {
-                  atomicAdd(&cache[ci].count, weight);
-                  cached = true;
-                  break;
-                }
-              }
-              ci = (ci + 1) & (LDS_CACHE - 1);
-            }
-            if (!cached) {
-              if (!agg_insert(A.tables[m], key, nk, weight))
-                atomicAdd(&lcnt[C_OVERFLOW], 1ull);
-            }
-          }
-        }
-      }
+      process_record(BV, start, end, A, P, fv, lcnt, cache, sig_lds,
+                     synth_lds);
     }
 
     if (staged) __syncthreads();  // tile reused next iteration
   }
+  }  // XP
 
   // flush LDS cache + counters
   __syncthreads();

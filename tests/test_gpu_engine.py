@@ -334,6 +334,41 @@ def test_literal_dotted_keys(engines, tmp_path):
         assert g.aggregators[0].ninputs == c.aggregators[0].ninputs
 
 
+def test_xpose_scan(engines, tmp_path):
+    """Wave-transposed staging (scan_kernel_x): length-sorted records
+    in 64B-granule interleaved layout must aggregate identically to
+    the oracle, including invalid/blank/edge lines."""
+    cpu, gpu = engines
+    from dragnet_amd.engine import plan as planmod
+    from dragnet_amd.engine.gpu import _ScanContext
+    from dragnet_amd.query import query_load
+    from dragnet_amd.tools.mktestdata import generate_lines
+    lines = list(generate_lines(30000, seed=99))
+    lines += [b"not json\n", b"\n", b'{"a": 1}\n',
+              b'{"req": {"method": "GET"}, "res": {"statusCode": 200}}\n',
+              b'{"m": "' + b"x" * 500 + b'"}\n']  # long record
+    pool = b"".join(lines)
+    path = tmp_path / "xp.ndjson"
+    path.write_bytes(pool)
+    for filt, spec in [({"eq": ["req.method", "GET"]},
+                        "req.method,res.statusCode"),
+                       (None, "latency[aggr=quantize]")]:
+        q = query_load(filter=filt, breakdown_specs=spec)
+        exp = cpu.scan([str(path)], [q])
+        cplan = planmod.compile_plan([q])
+        ctx = _ScanContext(gpu, cplan, agg_slots=1 << 15,
+                           dict_slots=1 << 15, dict_data_cap=8 << 20)
+        ctx.stage_xpose(pool)
+        for _ in range(2):  # repeat pass: reset correctness
+            ctx.reset()
+            ctx.scan_xpose()
+            aggs, stages = ctx.finalize([q])
+        assert aggs[0].points() == exp.aggregators[0].points(), \
+            (filt, spec)
+        assert dict(stages)["json parser"] == \
+            dict(exp.stages)["json parser"]
+
+
 def test_deep_nesting(engines, tmp_path):
     """Nesting to depth 64 parses identically to the oracle (the
     capture-slot window is 12 deep, but deeper containers sit inside

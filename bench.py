@@ -126,6 +126,9 @@ def main():
     # step); captured H2D memcpy nodes serialize against the kernels
     # and HURT the streaming path (35 vs 47 GB/s measured), so graphs
     # stay off when copies are in the loop.
+    # hipGraph replay pays on the linear device-resident path; it
+    # measured -6% on the wave-transposed path (single kernel + fused
+    # reset already) so xpose runs ungraphed.
     graph = None
     if (args.device_resident and not xpose
             and os.environ.get("DRAGNET_NO_GRAPH") != "1"):

@@ -591,21 +591,27 @@ class _ScanContext(object):
         self._scan_call(self._x["xdata"], 0, pos=rl, nlines=rl,
                         x=self._x)
 
-    def make_graph(self, h2d=True):
-        """Capture reset + the whole sliced scan pass into a hipGraph
-        so a step replays as one launch (the per-slice kernel+copy
-        launch overhead otherwise costs ~0.5-1 ms per step).  Returns
-        a replayable graph or None if capture fails."""
+    def make_graph(self, h2d=True, xpose=False):
+        """Capture reset + the whole scan pass into a hipGraph so a
+        step replays as one launch (the per-slice kernel+copy launch
+        overhead otherwise costs ~0.5-1 ms per step).  Returns a
+        replayable graph or None if capture fails."""
         torch = self.t
+
+        def pass_():
+            self.reset()
+            if xpose:
+                self.scan_xpose()
+            else:
+                self.scan_resident(h2d=h2d)
+
         try:
             # warm up the exact op sequence outside capture
-            self.reset()
-            self.scan_resident(h2d=h2d)
+            pass_()
             torch.cuda.synchronize(self.eng.device)
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
-                self.reset()
-                self.scan_resident(h2d=h2d)
+                pass_()
             torch.cuda.synchronize(self.eng.device)
             return g
         except Exception:

@@ -545,37 +545,7 @@ class _ScanContext(object):
         record length), fine for near-uniform record sizes."""
         torch = self.t
         dev = self.eng.device
-        arr = np.frombuffer(buf, dtype=np.uint8)
-        nl = np.flatnonzero(arr == 10).astype(np.int64)
-        if nl.size == 0:
-            raise ValueError("no records")
-        starts = np.empty_like(nl)
-        starts[0] = 0
-        starts[1:] = nl[:-1] + 1
-        lens = (nl - starts).astype(np.int64)
-        n = int(lens.size)
-        order = np.argsort(lens, kind="stable")
-        nslots = (n + 63) & ~63
-        g = max(1, (int(lens.max()) + 63) // 64)
-        K = g * 64
-        # gather each sorted record into a dense (n, K) matrix
-        so = starts[order]
-        lo = lens[order]
-        idx = so[:, None] + np.arange(K)[None, :]
-        np.minimum(idx, arr.size - 1, out=idx)
-        m = arr[idx]
-        m[np.arange(K)[None, :] >= lo[:, None]] = 10
-        if nslots > n:
-            m = np.concatenate(
-                [m, np.full((nslots - n, K), 10, dtype=np.uint8)])
-        nw = nslots // 64
-        # (wave, lane, granule, 64) -> (wave, granule, lane, 64)
-        xb = np.ascontiguousarray(
-            m.reshape(nw, 64, g, 64).transpose(0, 2, 1, 3)).reshape(-1)
-        xb = np.concatenate([xb, np.full(4096, 10, dtype=np.uint8)])
-        slot_len = np.full(nslots, 0xFFFFFFFF, dtype=np.uint32)
-        slot_len[:n] = lo.astype(np.uint32)
-        wave_base = (np.arange(nw, dtype=np.int64) * (g * 4096))
+        xb, wave_base, slot_len, nslots, n = _build_xpose_layout(buf)
         self._x = {
             "xdata": torch.from_numpy(xb).to(dev),
             "wave_base": torch.from_numpy(wave_base).to(dev),
@@ -767,6 +737,45 @@ class _ScanContext(object):
             "noutputs": 0,  # patched by caller if needed
             "nonnumeric": int(mc[CM_NONNUMERIC])}))
         return stages
+
+
+def _build_xpose_layout(buf):
+    """Numpy construction of the wave-transposed layout (see
+    _ScanContext.stage_xpose): returns (xbuf, wave_base, slot_len,
+    n_slots, n_records).  Byte p of slot r lives at
+    wave_base[r//64] + (p//64)*4096 + (r%64)*64 + p%64."""
+    arr = np.frombuffer(buf, dtype=np.uint8)
+    nl = np.flatnonzero(arr == 10).astype(np.int64)
+    if nl.size == 0:
+        raise ValueError("no records")
+    starts = np.empty_like(nl)
+    starts[0] = 0
+    starts[1:] = nl[:-1] + 1
+    lens = (nl - starts).astype(np.int64)
+    n = int(lens.size)
+    order = np.argsort(lens, kind="stable")
+    nslots = (n + 63) & ~63
+    g = max(1, (int(lens.max()) + 63) // 64)
+    K = g * 64
+    # gather each sorted record into a dense (n, K) matrix
+    so = starts[order]
+    lo = lens[order]
+    idx = so[:, None] + np.arange(K)[None, :]
+    np.minimum(idx, arr.size - 1, out=idx)
+    m = arr[idx]
+    m[np.arange(K)[None, :] >= lo[:, None]] = 10
+    if nslots > n:
+        m = np.concatenate(
+            [m, np.full((nslots - n, K), 10, dtype=np.uint8)])
+    nw = nslots // 64
+    # (wave, lane, granule, 64) -> (wave, granule, lane, 64)
+    xb = np.ascontiguousarray(
+        m.reshape(nw, 64, g, 64).transpose(0, 2, 1, 3)).reshape(-1)
+    xb = np.concatenate([xb, np.full(4096, 10, dtype=np.uint8)])
+    slot_len = np.full(nslots, 0xFFFFFFFF, dtype=np.uint32)
+    slot_len[:n] = lo.astype(np.uint32)
+    wave_base = (np.arange(nw, dtype=np.int64) * (g * 4096))
+    return xb, wave_base, slot_len, nslots, n
 
 
 def _decode_json_string(raw):

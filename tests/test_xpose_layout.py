@@ -1,0 +1,46 @@
+"""Wave-transposed layout builder (CPU): de-transposing every slot
+must reproduce the multiset of original records, pad slots are
+sentineled, and out-of-record bytes are newline fill (the device
+cursor may peek past a record's end)."""
+
+import numpy as np
+
+from dragnet_amd.engine.gpu import _build_xpose_layout
+
+
+def detranspose(xb, wave_base, slot_len, r):
+    length = int(slot_len[r])
+    w, lane = r // 64, r % 64
+    out = bytearray()
+    for p in range(length):
+        out.append(xb[int(wave_base[w]) + (p // 64) * 4096
+                      + lane * 64 + p % 64])
+    return bytes(out)
+
+
+def test_roundtrip_mixed_lengths():
+    recs = [b"{}", b'{"a": 1}', b"x" * 200, b"", b'{"b": [1,2,3]}',
+            b"y" * 70] * 30 + [b"z" * 513]
+    buf = b"\n".join(recs) + b"\n"
+    xb, wb, sl, nslots, n = _build_xpose_layout(buf)
+    assert n == len(recs)
+    assert nslots % 64 == 0 and nslots >= n
+    got = sorted(detranspose(xb, wb, sl, r) for r in range(n))
+    assert got == sorted(recs)
+    # pad slots carry the sentinel
+    for r in range(n, nslots):
+        assert sl[r] == 0xFFFFFFFF
+    # bytes past a record's end within its granule row are newline
+    r0 = next(r for r in range(n) if sl[r] == 0)
+    w, lane = r0 // 64, r0 % 64
+    assert xb[int(wb[w]) + lane * 64] == 10
+
+
+def test_single_wave_and_exact_multiple():
+    for count in (1, 64, 65, 128):
+        recs = [b"abc"] * count
+        buf = b"\n".join(recs) + b"\n"
+        xb, wb, sl, nslots, n = _build_xpose_layout(buf)
+        assert n == count and nslots == (count + 63) & ~63
+        for r in range(n):
+            assert detranspose(xb, wb, sl, r) == b"abc"

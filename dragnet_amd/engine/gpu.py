@@ -755,26 +755,39 @@ def _build_xpose_layout(buf):
     n = int(lens.size)
     order = np.argsort(lens, kind="stable")
     nslots = (n + 63) & ~63
-    g = max(1, (int(lens.max()) + 63) // 64)
-    K = g * 64
-    # gather each sorted record into a dense (n, K) matrix
+    nw = nslots // 64
     so = starts[order]
     lo = lens[order]
-    idx = so[:, None] + np.arange(K)[None, :]
-    np.minimum(idx, arr.size - 1, out=idx)
-    m = arr[idx]
-    m[np.arange(K)[None, :] >= lo[:, None]] = 10
-    if nslots > n:
-        m = np.concatenate(
-            [m, np.full((nslots - n, K), 10, dtype=np.uint8)])
-    nw = nslots // 64
-    # (wave, lane, granule, 64) -> (wave, granule, lane, 64)
-    xb = np.ascontiguousarray(
-        m.reshape(nw, 64, g, 64).transpose(0, 2, 1, 3)).reshape(-1)
-    xb = np.concatenate([xb, np.full(4096, 10, dtype=np.uint8)])
+    # per-wave granule count: lengths are sorted, so a wave's max is
+    # its last real lane; the granule-row stride is a constant 4096
+    # (64 lanes x 64B), so variable wave sizes need no device change
+    last = np.minimum(np.arange(nw) * 64 + 63, n - 1)
+    gw = np.maximum(1, (lo[last] + 63) // 64).astype(np.int64)
+    wave_base = np.zeros(nw, dtype=np.int64)
+    np.cumsum(gw[:-1] * 4096, out=wave_base[1:])
+    total = int(wave_base[-1] + gw[-1] * 4096) + 4096  # +slack row
+    xb = np.full(total, 10, dtype=np.uint8)
+    # build waves in batches of equal granule count (few distinct
+    # values after sorting) with one vectorized gather per batch
+    for g in np.unique(gw):
+        ws = np.flatnonzero(gw == g)
+        K = int(g) * 64
+        rsel = (ws[:, None] * 64 + np.arange(64)[None, :]).reshape(-1)
+        rl = np.where(rsel < n, lo[np.minimum(rsel, n - 1)], 0)
+        rs = np.where(rsel < n, so[np.minimum(rsel, n - 1)], 0)
+        idx = rs[:, None] + np.arange(K)[None, :]
+        np.minimum(idx, arr.size - 1, out=idx)
+        m = arr[idx]
+        m[np.arange(K)[None, :] >= rl[:, None]] = 10
+        # (wave, lane, granule, 64) -> (wave, granule, lane, 64)
+        blk = np.ascontiguousarray(
+            m.reshape(len(ws), 64, int(g), 64).transpose(0, 2, 1, 3)
+        ).reshape(len(ws), -1)
+        for i, w in enumerate(ws):
+            b = int(wave_base[w])
+            xb[b:b + K * 64] = blk[i]
     slot_len = np.full(nslots, 0xFFFFFFFF, dtype=np.uint32)
     slot_len[:n] = lo.astype(np.uint32)
-    wave_base = (np.arange(nw, dtype=np.int64) * (g * 4096))
     return xb, wave_base, slot_len, nslots, n
 
 

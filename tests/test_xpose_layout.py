@@ -36,6 +36,18 @@ def test_roundtrip_mixed_lengths():
     assert xb[int(wb[w]) + lane * 64] == 10
 
 
+def test_skewed_lengths_compact():
+    """Per-wave granule counts: one 8KB record must not inflate every
+    wave (was a global-max prototype limitation)."""
+    recs = [b"s" * 100] * 1000 + [b"L" * 8000]
+    buf = b"\n".join(recs) + b"\n"
+    xb, wb, sl, nslots, n = _build_xpose_layout(buf)
+    # compact bound: ~2 granules per short record + the one big wave
+    assert xb.size < 1000 * 2 * 64 * 2 + 64 * 8064 + 8192
+    got = sorted(detranspose(xb, wb, sl, r) for r in range(n))
+    assert got == sorted(recs)
+
+
 def test_single_wave_and_exact_multiple():
     for count in (1, 64, 65, 128):
         recs = [b"abc"] * count

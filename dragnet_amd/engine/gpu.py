@@ -545,7 +545,9 @@ class _ScanContext(object):
         record length), fine for near-uniform record sizes."""
         torch = self.t
         dev = self.eng.device
-        xb, wave_base, slot_len, nslots, n = _build_xpose_layout(buf)
+        gran = _env_int("DRAGNET_XGRAN", 32)
+        xb, wave_base, slot_len, nslots, n = _build_xpose_layout(
+            buf, gran)
         self._x = {
             "xdata": torch.from_numpy(xb).to(dev),
             "wave_base": torch.from_numpy(wave_base).to(dev),
@@ -739,11 +741,12 @@ class _ScanContext(object):
         return stages
 
 
-def _build_xpose_layout(buf):
+def _build_xpose_layout(buf, gran=64):
     """Numpy construction of the wave-transposed layout (see
     _ScanContext.stage_xpose): returns (xbuf, wave_base, slot_len,
     n_slots, n_records).  Byte p of slot r lives at
-    wave_base[r//64] + (p//64)*4096 + (r%64)*64 + p%64."""
+    wave_base[r//64] + (p//gran)*(64*gran) + (r%64)*gran + p%gran."""
+    stride = 64 * gran
     arr = np.frombuffer(buf, dtype=np.uint8)
     nl = np.flatnonzero(arr == 10).astype(np.int64)
     if nl.size == 0:
@@ -762,16 +765,16 @@ def _build_xpose_layout(buf):
     # its last real lane; the granule-row stride is a constant 4096
     # (64 lanes x 64B), so variable wave sizes need no device change
     last = np.minimum(np.arange(nw) * 64 + 63, n - 1)
-    gw = np.maximum(1, (lo[last] + 63) // 64).astype(np.int64)
+    gw = np.maximum(1, (lo[last] + gran - 1) // gran).astype(np.int64)
     wave_base = np.zeros(nw, dtype=np.int64)
-    np.cumsum(gw[:-1] * 4096, out=wave_base[1:])
-    total = int(wave_base[-1] + gw[-1] * 4096) + 4096  # +slack row
+    np.cumsum(gw[:-1] * stride, out=wave_base[1:])
+    total = int(wave_base[-1] + gw[-1] * stride) + stride  # +slack
     xb = np.full(total, 10, dtype=np.uint8)
     # build waves in batches of equal granule count (few distinct
     # values after sorting) with one vectorized gather per batch
     for g in np.unique(gw):
         ws = np.flatnonzero(gw == g)
-        K = int(g) * 64
+        K = int(g) * gran
         rsel = (ws[:, None] * 64 + np.arange(64)[None, :]).reshape(-1)
         rl = np.where(rsel < n, lo[np.minimum(rsel, n - 1)], 0)
         rs = np.where(rsel < n, so[np.minimum(rsel, n - 1)], 0)
@@ -781,7 +784,7 @@ def _build_xpose_layout(buf):
         m[np.arange(K)[None, :] >= rl[:, None]] = 10
         # (wave, lane, granule, 64) -> (wave, granule, lane, 64)
         blk = np.ascontiguousarray(
-            m.reshape(len(ws), 64, int(g), 64).transpose(0, 2, 1, 3)
+            m.reshape(len(ws), 64, int(g), gran).transpose(0, 2, 1, 3)
         ).reshape(len(ws), -1)
         for i, w in enumerate(ws):
             b = int(wave_base[w])

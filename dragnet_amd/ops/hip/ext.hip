@@ -199,15 +199,25 @@ void scan_chunk(
   if (A.xn_slots > 0) {
     blocks = (A.xn_slots + BLOCK - 1) / BLOCK;
     if (blocks > 2048) blocks = 2048;
-    if (mw == 2)
-      hipLaunchKernelGGL(scan_kernel_x<2>, dim3(blocks), dim3(BLOCK),
-                         lds, current_stream(), A);
+    // 32B granules measured fastest (less padding, tighter rows):
+    // 1101 vs 1079 (64B) vs 1023 (128B) M rec/s on one box
+    const char* xg_env = getenv("DRAGNET_XGRAN");
+    int xg = xg_env ? atoi(xg_env) : 32;
+    if (xg == 32)
+      hipLaunchKernelGGL((scan_kernel_x<4, 5>), dim3(blocks),
+                         dim3(BLOCK), lds, current_stream(), A);
+    else if (xg == 128)
+      hipLaunchKernelGGL((scan_kernel_x<4, 7>), dim3(blocks),
+                         dim3(BLOCK), lds, current_stream(), A);
+    else if (mw == 2)
+      hipLaunchKernelGGL((scan_kernel_x<2, 6>), dim3(blocks),
+                         dim3(BLOCK), lds, current_stream(), A);
     else if (mw == 3)
-      hipLaunchKernelGGL(scan_kernel_x<3>, dim3(blocks), dim3(BLOCK),
-                         lds, current_stream(), A);
+      hipLaunchKernelGGL((scan_kernel_x<3, 6>), dim3(blocks),
+                         dim3(BLOCK), lds, current_stream(), A);
     else
-      hipLaunchKernelGGL(scan_kernel_x<4>, dim3(blocks), dim3(BLOCK),
-                         lds, current_stream(), A);
+      hipLaunchKernelGGL((scan_kernel_x<4, 6>), dim3(blocks),
+                         dim3(BLOCK), lds, current_stream(), A);
     hipError_t xerr = hipGetLastError();
     TORCH_CHECK(xerr == hipSuccess, "scan_kernel_x launch failed: ",
                 hipGetErrorString(xerr));

@@ -130,26 +130,26 @@ struct Bytes {
 // layout's 64 scattered records (the SQ_WAIT-bound gather profile).
 // Positions are RECORD-RELATIVE (start=0); spans captured from this
 // source are decoded through the same XBytes.
-constexpr uint32_t XGRAN = 64;            // granule bytes
-constexpr uint32_t XSTRIDE = 64 * XGRAN;  // wave bytes per granule row
-
-struct XBytes {
-  const uint8_t* base;  // pool + wave_base + lane*XGRAN
+template <int LG>  // log2 granule bytes (5/6/7 = 32/64/128B)
+struct XBytesT {
+  static constexpr uint32_t GRAN = 1u << LG;
+  static constexpr uint32_t STRIDE = 64u << LG;  // wave granule row
+  const uint8_t* base;  // pool + wave_base + lane*GRAN
   DEV uint8_t at(uint32_t p) const {
-    return base[(size_t)(p >> 6) * XSTRIDE + (p & 63u)];
+    return base[(size_t)(p >> LG) * STRIDE + (p & (GRAN - 1u))];
   }
   DEV uint64_t load8(uint32_t p) const {
-    uint32_t o = p & 63u;
-    const uint8_t* gp = base + (size_t)(p >> 6) * XSTRIDE;
+    uint32_t o = p & (GRAN - 1u);
+    const uint8_t* gp = base + (size_t)(p >> LG) * STRIDE;
     uint64_t w;
-    if (o <= XGRAN - 8) {
+    if (o <= GRAN - 8) {
       __builtin_memcpy(&w, gp + o, 8);
       return w;
     }
     uint64_t lo, hi;
-    __builtin_memcpy(&lo, gp + XGRAN - 8, 8);
-    __builtin_memcpy(&hi, gp + XSTRIDE, 8);
-    uint32_t sh = (o - (XGRAN - 8)) * 8;  // 8..56
+    __builtin_memcpy(&lo, gp + GRAN - 8, 8);
+    __builtin_memcpy(&hi, gp + STRIDE, 8);
+    uint32_t sh = (o - (GRAN - 8)) * 8;  // 8..56
     return (lo >> sh) | (hi << (64 - sh));
   }
 };
@@ -1329,16 +1329,19 @@ template __global__ void scan_kernel_mw<4>(ScanArgs);
 template __global__ void scan_kernel_mw<5>(ScanArgs);
 template __global__ void scan_kernel_mw<6>(ScanArgs);
 
-// wave-transposed record staging (XBytes; DRAGNET_XPOSE)
-template <int MW>
+// wave-transposed record staging (XBytesT; DRAGNET_XPOSE).
+// second parameter = log2 granule bytes (DRAGNET_XGRAN)
+template <int MW, int XLG>
 __launch_bounds__(BLOCK, MW)
 __global__ void scan_kernel_x(ScanArgs A) {
   extern __shared__ __attribute__((aligned(16))) char smem2[];
-  scan_kernel_body<1>(smem2, A);
+  scan_kernel_body<XLG>(smem2, A);
 }
-template __global__ void scan_kernel_x<2>(ScanArgs);
-template __global__ void scan_kernel_x<3>(ScanArgs);
-template __global__ void scan_kernel_x<4>(ScanArgs);
+template __global__ void scan_kernel_x<2, 6>(ScanArgs);
+template __global__ void scan_kernel_x<3, 6>(ScanArgs);
+template __global__ void scan_kernel_x<4, 6>(ScanArgs);
+template __global__ void scan_kernel_x<4, 5>(ScanArgs);
+template __global__ void scan_kernel_x<4, 7>(ScanArgs);
 
 // Per-record pipeline (K1-K6) shared by the linear and the
 // wave-transposed (XBytes) kernels: tokenize/extract, predicate,
@@ -1632,12 +1635,13 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
   fv.type = fv_type; fv.soff = fv_soff; fv.slen = fv_slen; fv.num = fv_num;
   fv.tid = threadIdx.x;
 
-  if constexpr (XP == 1) {
-    // wave-transposed pool: slot r's bytes live granule-interleaved
-    // at xwave_base[r/64] + (r%64)*XGRAN; a wave's refills are 64
-    // consecutive granules (coalesced) instead of 64 scattered
-    // records.  Record order is length-sorted by the host — the
-    // aggregation is order-independent (associative merge).
+  if constexpr (XP != 0) {
+    // wave-transposed pool (XP = log2 granule bytes): slot r's bytes
+    // live granule-interleaved at xwave_base[r/64] + (r%64)*GRAN; a
+    // wave's refills are 64 consecutive granules (coalesced) instead
+    // of 64 scattered records.  Record order is length-sorted by the
+    // host — the aggregation is order-independent (associative
+    // merge).
     const uint32_t stride_x = gridDim.x * BLOCK;
     for (uint32_t rbase = blockIdx.x * BLOCK; rbase < A.xn_slots;
          rbase += stride_x) {
@@ -1645,9 +1649,9 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
       if (r < A.xn_slots) {
         uint32_t len = A.xrec_len[r];
         if (len != 0xFFFFFFFFu) {
-          XBytes BV;
+          XBytesT<XP> BV;
           BV.base = A.xdata + A.xwave_base[r >> 6]
-                    + (size_t)(r & 63u) * XGRAN;
+                    + ((size_t)(r & 63u) << XP);
           process_record(BV, 0u, len, A, P, fv, lcnt, cache, sig_lds,
                          synth_lds);
         }

@@ -8,13 +8,13 @@ import numpy as np
 from dragnet_amd.engine.gpu import _build_xpose_layout
 
 
-def detranspose(xb, wave_base, slot_len, r):
+def detranspose(xb, wave_base, slot_len, r, gran=64):
     length = int(slot_len[r])
     w, lane = r // 64, r % 64
     out = bytearray()
     for p in range(length):
-        out.append(xb[int(wave_base[w]) + (p // 64) * 4096
-                      + lane * 64 + p % 64])
+        out.append(xb[int(wave_base[w]) + (p // gran) * (64 * gran)
+                      + lane * gran + p % gran])
     return bytes(out)
 
 
@@ -56,3 +56,13 @@ def test_single_wave_and_exact_multiple():
         assert n == count and nslots == (count + 63) & ~63
         for r in range(n):
             assert detranspose(xb, wb, sl, r) == b"abc"
+
+
+def test_granule_sizes():
+    recs = [b"g" * 40, b"h" * 100, b"i" * 200] * 50
+    buf = b"\n".join(recs) + b"\n"
+    for gran in (32, 64, 128):
+        xb, wb, sl, nslots, n = _build_xpose_layout(buf, gran)
+        got = sorted(detranspose(xb, wb, sl, r, gran)
+                     for r in range(n))
+        assert got == sorted(recs), gran

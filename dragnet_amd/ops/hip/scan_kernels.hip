@@ -124,12 +124,12 @@ struct Bytes {
 
 // Wave-transposed byte source (DRAGNET_XPOSE resident staging): the
 // pool is re-laid out so granule g of the record owned by lane l of a
-// wave sits at wave_base + g*(64*XGRAN) + l*XGRAN.  A wave's 64
-// window refills for granule g then touch 64 *consecutive* XGRAN-byte
-// blocks — fully coalesced, ~4x fewer cache lines than the linear
-// layout's 64 scattered records (the SQ_WAIT-bound gather profile).
-// Positions are RECORD-RELATIVE (start=0); spans captured from this
-// source are decoded through the same XBytes.
+// wave sits at wave_base + g*(64*GRAN) + l*GRAN.  A wave's window
+// refills for granule g then touch 64 *consecutive* GRAN-byte blocks
+// — coalesced, several times fewer cache lines than the linear
+// layout's 64 scattered records (the SQ_WAIT-bound gather profile in
+// profiles/).  Positions are RECORD-RELATIVE (start=0); spans
+// captured from this source are decoded through the same XBytesT.
 template <int LG>  // log2 granule bytes (5/6/7 = 32/64/128B)
 struct XBytesT {
   static constexpr uint32_t GRAN = 1u << LG;

@@ -469,7 +469,13 @@ def _output_result(query, opts, result, title=None):
 
     if opts.get("counters"):
         stages = list(result.stages)
-        stages.append(("Flattener", {"ninputs": npoints, "noutputs": 1}))
+        if not opts.get("points"):
+            # the flattener stage exists only when output is
+            # re-aggregated to rows; --points streams raw points
+            # (reference golden: tst.scan_fileset.sh.out --points
+            # --counters section has no Flattener lines)
+            stages.append(("Flattener",
+                           {"ninputs": npoints, "noutputs": 1}))
         mod_output.dump_counters(stages)
 
     if opts.get("warnings"):

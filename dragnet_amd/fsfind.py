@@ -17,6 +17,10 @@ import stat as _stat
 
 class FindCounters(object):
     def __init__(self):
+        self.npathenum = None  # noutputs incl. the EOF push, when
+                               # strftime pruning ran (vstream counts
+                               # the stream-end: tst.scan_fileset
+                               # golden shows npaths+1)
         self.nstarts = 0
         self.nstatted = 0
         self.ndirectories = 0
@@ -25,14 +29,25 @@ class FindCounters(object):
         self.noutputs = 0
 
     def stages(self):
-        return [
+        # The statter/traverser also pass the generation-numbered EOF
+        # markers the feedback loop cycles (one per expanded directory
+        # plus one per start; reference lib/fs-find.js:185-205) —
+        # pinned by the reference goldens: tst.empty.sh.out statter
+        # ninputs 2 for one char device, tst.scan_fileset.sh.out 24
+        # for 16 paths + 7 directories + 1 start.
+        cycled = self.nstatted + self.ndirectories + self.nstarts
+        head = []
+        if self.npathenum is not None:
+            head.append(("PathEnumerator",
+                         {"noutputs": self.npathenum}))
+        return head + [
             ("FindStart", {"ninputs": self.nstarts,
                            "noutputs": self.nstarts}),
-            ("FindStatter", {"ninputs": self.nstatted,
-                             "noutputs": self.nstatted}),
-            ("FindTraverser", {"ninputs": self.nstatted,
-                               "noutputs": self.nstatted}),
-            ("FindFeedback", {"ninputs": self.nstatted,
+            ("FindStatter", {"ninputs": cycled,
+                             "noutputs": cycled}),
+            ("FindTraverser", {"ninputs": cycled,
+                               "noutputs": cycled}),
+            ("FindFeedback", {"ninputs": cycled,
                               "noutputs": self.noutputs,
                               "ndirectories": self.ndirectories,
                               "nregfiles": self.nregfiles,
@@ -88,7 +103,9 @@ def find_data_files(root, timeformat=None, after_ms=None, before_ms=None,
     """
     from . import pathenum
     if before_ms is not None and timeformat is not None:
-        pats = pathenum.enumerate_paths(
-            os.path.join(root, timeformat), after_ms, before_ms)
+        pats = list(pathenum.enumerate_paths(
+            os.path.join(root, timeformat), after_ms, before_ms))
+        if counters is not None:
+            counters.npathenum = len(pats) + 1
         return find_files(pats, counters=counters, warn=warn)
     return find_files([root], counters=counters, warn=warn)

@@ -1468,6 +1468,20 @@ DEV void process_record(BS BV, uint32_t start, uint32_t end,
           }
           uint32_t code;
           if (B[2] != BUCKET_NONE) {
+            if (t == T_STR) {
+              // numeric STRINGS coerce for bucketized fields (JS
+              // arithmetic in the reference's bucketizer: its own
+              // golden counts {"latency": "26"} into the p2
+              // histogram — tests/data/2014/05-05/more.log:1);
+              // NaN / +-Infinity drop as nonnumeric
+              num = js_to_number(BV, so, sl);
+              if (!(num == num) || num == __builtin_inf() ||
+                  num == -__builtin_inf()) {
+                drop = true;
+                break;
+              }
+              t = T_NUM;
+            }
             if (t != T_NUM) { drop = true; break; }  // nonnumeric
             long long ord;
             if (B[2] == BUCKET_P2) {

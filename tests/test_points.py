@@ -91,11 +91,16 @@ def test_aggregator_quantize_drop_nonnumeric():
     q = query_load(breakdown_specs="latency[aggr=quantize]")
     a = Aggregator(q)
     a.write({"fields": {"latency": 5}, "value": 1})
-    a.write({"fields": {"latency": "5"}, "value": 1})   # string: dropped
+    # numeric STRINGS coerce (JS arithmetic in the reference's
+    # bucketizer; proved by its own fileset golden counting
+    # {"latency": "26"} — despite README.md:718-722's claim)
+    a.write({"fields": {"latency": "5"}, "value": 1})
+    a.write({"fields": {"latency": "abc"}, "value": 1})  # NaN: dropped
+    a.write({"fields": {"latency": "Infinity"}, "value": 1})  # dropped
     a.write({"fields": {"latency": True}, "value": 1})  # bool: dropped
     a.write({"fields": {}, "value": 1})                 # missing: dropped
-    assert a.ndropped_nonnumeric == 3
-    assert a.points() == [{"fields": {"latency": 4}, "value": 1}]
+    assert a.ndropped_nonnumeric == 4
+    assert a.points() == [{"fields": {"latency": 4}, "value": 2}]
 
 
 def test_aggregator_zero_breakdowns():

@@ -78,7 +78,8 @@ def check_conservation(stages):
     for name, counters in stages:
         if "ninputs" not in counters or "noutputs" not in counters:
             continue
-        if name in ("Aggregator", "Flattener", "IndexQuery") \
+        if name in ("Aggregator", "Flattener",
+                    "Index Result Aggregator") \
                 or name.startswith("Find"):
             continue  # aggregation compresses; finders emit subsets
         drops = sum(counters.get(k, 0) for k in drop_keys)

@@ -9,6 +9,7 @@ points (reference bin/dn:941-1274).
 """
 
 import json
+import os
 
 import sys
 
@@ -22,7 +23,7 @@ from .query import QueryConfig, QueryError
 USAGE = """\
 usage: dn SUBCOMMAND [OPTIONS] ARGS
 
-dn datasource-add    [--backend=file|sharded] --path=DATA_PATH
+dn datasource-add    [--backend=file|sharded|manta] --path=DATA_PATH
                      [--index-path=INDEX_PATH] [--filter=FILTER]
                      [--time-field=FIELD] [--time-format=TIME_FORMAT]
                      [--data-format=json|json-skinner] [--shards=N]
@@ -93,7 +94,7 @@ def main(argv=None):
 
 def _dispatch(argv):
     if not argv:
-        return usage("no command specified")
+        return usage("no command specified", full=True)
 
     cmdname = argv[0]
     cmds = {
@@ -128,9 +129,15 @@ def _dispatch(argv):
         return 0
 
 
-def usage(msg):
+def usage(msg, full=False):
+    """Error + short usage line (the reference goldens carry exactly
+    these two lines on errors, tst.config.sh.out:7-9); the full
+    option table prints only for a bare `dn` (full=True)."""
     sys.stderr.write("dn: %s\n" % msg)
-    sys.stderr.write(USAGE)
+    if full:
+        sys.stderr.write(USAGE)
+    else:
+        sys.stderr.write("usage: dn SUBCOMMAND [OPTIONS] ARGS\n")
     return 2
 
 
@@ -214,7 +221,12 @@ def parse_args(argv, allowed):
             args.append(a)
             i += 1
             continue
-        opts[name] = val if hasval else True
+        if name == "breakdowns" and "breakdowns" in opts:
+            # repeated -b accumulates (reference dashdash
+            # arrayOfString; tst.index_fileset.sh passes two -b)
+            opts[name] = opts[name] + "," + val
+        else:
+            opts[name] = val if hasval else True
         i += 1
     return opts, args
 
@@ -341,11 +353,18 @@ def cmd_datasource_remove(argv):
 def _ds_location(ds):
     if ds.backend == "sharded":
         return "sharded:/" + (ds.path or "")
+    if ds.backend == "manta":
+        # reference renders the Manta host from MANTA_URL
+        # (default us-east.manta.joyent.com)
+        url = os.environ.get("MANTA_URL",
+                             "https://us-east.manta.joyent.com")
+        host = url.split("://", 1)[-1].rstrip("/")
+        return "manta://" + host + (ds.path or "")
     return "file:/" + (ds.path or "")
 
 
 def _print_ds(out, ds, verbose):
-    out.write("%-20s %-58s\n" % (ds.name, _ds_location(ds)))
+    out.write("%-20s %-59s\n" % (ds.name, _ds_location(ds)))
     if not verbose:
         return
     if ds.filter is not None:
@@ -370,7 +389,7 @@ def cmd_datasource_list(argv):
     check_arg_count(args, 0)
     cfg = _load_config()
     out = sys.stdout
-    out.write("%-20s %-58s\n" % ("DATASOURCE", "LOCATION"))
+    out.write("%-20s %-59s\n" % ("DATASOURCE", "LOCATION"))
     for ds in cfg.datasource_list():
         _print_ds(out, ds, opts.get("verbose", False))
 
@@ -381,7 +400,7 @@ def cmd_datasource_show(argv):
     cfg = _load_config()
     ds = _get_datasource(cfg, args[0])
     out = sys.stdout
-    out.write("%-20s %-58s\n" % ("DATASOURCE", "LOCATION"))
+    out.write("%-20s %-59s\n" % ("DATASOURCE", "LOCATION"))
     _print_ds(out, ds, opts.get("verbose", False))
 
 

@@ -13,7 +13,11 @@ import os
 CONFIG_MAJOR = 0
 CONFIG_MINOR = 0
 
-VALID_BACKENDS = ("file", "sharded")
+# "manta" is accepted at the CONFIG layer for reference parity (the
+# reference stores manta datasources without validating reachability,
+# tst.config.sh); USING one errors — the sharded backend is this
+# framework's distributed analog.
+VALID_BACKENDS = ("file", "sharded", "manta")
 VALID_FORMATS = ("json", "json-skinner")
 
 
@@ -137,7 +141,9 @@ class DragnetConfig(object):
             if m.datasource == dsname and m.name == name:
                 del self.metrics[i]
                 return
-        raise ConfigError('metric "%s" does not exist' % name)
+        raise ConfigError(
+            'datasource "%s" metric "%s" does not exist'
+            % (dsname, name))
 
     def datasource_metrics(self, dsname):
         return [m for m in self.metrics if m.datasource == dsname]

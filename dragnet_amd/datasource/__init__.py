@@ -9,4 +9,8 @@ def datasource_for_config(ds, engine=None):
     if ds.backend == "sharded":
         from .sharded import ShardedDatasource
         return ShardedDatasource(ds, engine=engine)
+    if ds.backend == "manta":
+        raise ValueError(
+            'the "manta" backend is not available in this deployment; '
+            'use the "sharded" backend for distributed scans')
     raise ValueError('unsupported backend: "%s"' % ds.backend)

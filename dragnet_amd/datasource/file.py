@@ -202,14 +202,14 @@ class FileDatasource(object):
 
     # ---- index querying ----
 
-    def _find_index_files(self, query, interval):
+    def _find_index_files(self, query, interval, counters=None):
         if interval == "all":
             root = os.path.join(self.ds.index_path, "all")
-            return find_files([root])
+            return find_files([root], counters=counters)
         _, _, subdir, pattern, _ = INTERVALS[interval]
         root = os.path.join(self.ds.index_path, subdir)
         return find_data_files(root, pattern, query.after_ms,
-                               query.before_ms)
+                               query.before_ms, counters=counters)
 
     def query(self, query, interval="day", dry_run=False, out=None):
         """Answer a query from the index tree: per-file partials merged
@@ -218,7 +218,9 @@ class FileDatasource(object):
         if not self.ds.index_path:
             raise ValueError(
                 'datasource is missing "indexPath" for index operations')
-        files = list(self._find_index_files(query, interval))
+        fcounters = FindCounters()
+        files = list(self._find_index_files(query, interval,
+                                            counters=fcounters))
         if dry_run:
             out = out or sys.stderr
             out.write("would scan files:\n")
@@ -257,8 +259,20 @@ class FileDatasource(object):
                 nerrors.append((path, str(e)))
             finally:
                 iq.close()
-        stages = [("IndexQuery", {
-            "nfiles": len(files), "nerrors": len(nerrors)})]
+        # Counter stages mirror the reference's query pipeline
+        # (lib/datasource-file.js:608-641): the find pipeline over the
+        # index tree, then the "Index List" passthrough and the result
+        # aggregator — BOTH carry the partial result rows streamed out
+        # of the per-file index queriers (pinned across four goldens:
+        # 2 rows/1 file, 24 rows/24 files, 120 rows/120 files, 1
+        # row/empty index).
+        stages = fcounters.stages() + [
+            ("Index List", {"ninputs": final.ninputs,
+                            "noutputs": final.ninputs}),
+            ("Index Result Aggregator",
+             {"ninputs": final.ninputs,
+              "noutputs": final.noutputs()}),
+        ]
         result = ScanResult([final], stages, [p for p, _ in files])
         result.errors = nerrors
         return result
@@ -338,9 +352,9 @@ def _esc(c):
     return sqlite3_escape(c)
 
 
-def find_files(roots):
+def find_files(roots, counters=None):
     from ..fsfind import find_files as ff
-    return ff(roots)
+    return ff(roots, counters=counters)
 
 
 def write_index(index_path, metrics, interval, points):

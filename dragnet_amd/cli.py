@@ -221,10 +221,11 @@ def parse_args(argv, allowed):
             args.append(a)
             i += 1
             continue
-        if name == "breakdowns" and "breakdowns" in opts:
-            # repeated -b accumulates (reference dashdash
-            # arrayOfString; tst.index_fileset.sh passes two -b)
-            opts[name] = opts[name] + "," + val
+        if name == "breakdowns":
+            # repeated -b accumulates as a LIST (reference dashdash
+            # arrayOfString): each entry is attr-parsed separately so
+            # errors quote only the offending value
+            opts.setdefault(name, []).append(val)
         else:
             opts[name] = val if hasval else True
         i += 1
@@ -248,10 +249,14 @@ def _parse_filter(s):
         if "Expecting" in msg and "char 0" not in msg:
             msg = "Unexpected end of input" \
                 if "line" in msg and s.strip() in ("{", "[") else msg
-        raise FatalError("invalid filter: %s" %
+        # a usage-class error: the reference routes JSON parse
+        # failures through usage() (bin/dn dnParseArgs)
+        raise UsageError("invalid filter: %s" %
                          ("Unexpected end of input"
                           if _is_truncated_json(s) else msg))
-    krill.create_predicate(f)  # validate
+    # predicate semantics are validated where the query is built
+    # (QueryConfig), so errors carry the reference's full chain:
+    # "invalid query: invalid filter: predicate ...: unknown operator"
     return f
 
 
@@ -300,10 +305,11 @@ def _datasource_from_opts(name, opts, existing=None):
         filt = _parse_filter(filt)
     elif existing is not None:
         filt = existing.filter
+    # the data format is validated at USE time, not add time (the
+    # reference stores it and `dn scan` fails with
+    # 'unsupported format: "junk"' — tst.badargs.sh)
     data_format = opts.get(
         "data-format", existing.data_format if existing else "json")
-    if data_format not in mod_config.VALID_FORMATS:
-        raise FatalError('unsupported data format: "%s"' % data_format)
     nshards = opts.get("shards", existing.nshards if existing else None)
     if nshards is not None:
         nshards = int(nshards)
@@ -406,6 +412,19 @@ def cmd_datasource_show(argv):
 
 # ---- metric commands ----
 
+def _parse_breakdowns(values):
+    """attr-parse each -b occurrence separately (reference
+    dnExpandArray, bin/dn; errors quote the offending entry)."""
+    parsed = []
+    for v in values:
+        one = attrs_parse(v)
+        if isinstance(one, AttrsError):
+            raise UsageError(
+                'bad value for "breakdowns" ("%s"): %s' % (v, one))
+        parsed.extend(one)
+    return parsed
+
+
 def cmd_metric_add(argv):
     opts, args = parse_args(argv, ["filter", "breakdowns"])
     check_arg_count(args, 2)
@@ -415,11 +434,8 @@ def cmd_metric_add(argv):
     filt = _parse_filter(opts.get("filter"))
     breakdowns = []
     if opts.get("breakdowns"):
-        parsed = attrs_parse(opts["breakdowns"])
-        if isinstance(parsed, AttrsError):
-            raise FatalError("invalid breakdowns: %s" % parsed)
         from .query import parse_fields
-        breakdowns = parse_fields(parsed)
+        breakdowns = parse_fields(_parse_breakdowns(opts["breakdowns"]))
     cfg.metric_add(mod_config.Metric(
         name=metname, datasource=dsname, filter=filt,
         breakdowns=breakdowns))
@@ -459,17 +475,20 @@ def _query_from_opts(opts, allow_reserved=False):
     filt = _parse_filter(opts.get("filter"))
     breakdowns = []
     if opts.get("breakdowns"):
-        parsed = attrs_parse(opts["breakdowns"])
-        if isinstance(parsed, AttrsError):
-            raise FatalError("invalid breakdowns: %s" % parsed)
-        breakdowns = parsed
+        breakdowns = _parse_breakdowns(opts["breakdowns"])
     try:
-        return QueryConfig(
+        qc = QueryConfig(
             filter=filt, breakdowns=breakdowns,
             time_after=opts.get("after"), time_before=opts.get("before"),
             allow_reserved=allow_reserved)
     except QueryError as e:
         raise FatalError("invalid query: %s" % e)
+    if opts.get("gnuplot") and len(qc.breakdowns) != 1:
+        # validated up front, before any scanning
+        # (reference bin/dn:713-716)
+        raise FatalError(
+            "--gnuplot can only be used with exactly one breakdown")
+    return qc
 
 
 def _output_result(query, opts, result, title=None):

@@ -56,7 +56,8 @@ def _validate(pred):
             _validate(sub)
         return
     if key not in LEAF_OPS:
-        raise KrillError('unknown operator: "%s"' % key)
+        raise KrillError('predicate %s: unknown operator "%s"'
+                         % (js_inspect(pred), key))
     if (not isinstance(val, list) or len(val) != 2
             or not isinstance(val[0], str)):
         raise KrillError('"%s" requires [fieldname, value]' % key)
@@ -145,6 +146,39 @@ def _relational(a, b, op):
     if op == "gt":
         return x > y
     return x >= y
+
+
+def js_inspect(obj):
+    """Node util.inspect rendering of a predicate (the reference's
+    error messages embed it: tst.badargs.sh.out:9)."""
+    if obj is None:
+        return "null"
+    if obj is True:
+        return "true"
+    if obj is False:
+        return "false"
+    if isinstance(obj, str):
+        return "'" + obj.replace("\\", "\\\\").replace("'", "\\'") + "'"
+    if isinstance(obj, (int, float)):
+        return js_num_repr(obj)
+    if isinstance(obj, list):
+        if not obj:
+            return "[]"
+        return "[ " + ", ".join(js_inspect(x) for x in obj) + " ]"
+    if isinstance(obj, dict):
+        if not obj:
+            return "{}"
+        return "{ " + ", ".join(
+            "%s: %s" % (k, js_inspect(v)) for k, v in obj.items()) + " }"
+    return repr(obj)
+
+
+def js_num_repr(v):
+    if isinstance(v, bool):
+        return "true" if v else "false"
+    if isinstance(v, int) or (isinstance(v, float) and v.is_integer()):
+        return str(int(v))
+    return repr(v)
 
 
 def pluck(fields, path):

@@ -415,6 +415,45 @@ def test_config_goldens(dn):
     assert ran >= 45
 
 
+def test_badargs_goldens(dn):
+    """tst.badargs.sh.out: error wording for malformed breakdowns,
+    truncated/unknown-operator filters (node util.inspect predicate
+    rendering), --gnuplot arity, and an unsupported data format
+    validated at USE time.  The harness pipes through `head -2`, so
+    each case pins the first two merged-output lines."""
+    one = os.path.join(DATA, "2014", "05-01", "one.log")
+    golden = open(os.path.join(
+        REF, "dn", "local", "tst.badargs.sh.out")).read().splitlines()
+    assert dn("datasource-add", "--path=" + one, "input").code == 0
+
+    cases = [
+        ["scan", "-b", "host", "-b", "req.method,x[=bar]", "input"],
+        ["scan", "-b", "host", "-b", "req.method,[]", "input"],
+        ["scan", "-b", "host", "-b", "req.method,foo[", "input"],
+        ["scan", "-f", "{", "input"],
+        ["scan", "-f", '{ "junk": [ "foo", "bar" ] }', "input"],
+        ["scan", "--gnuplot", "input"],
+        ["scan", "-b", "req.method,res.statusCode", "--gnuplot",
+         "input"],
+    ]
+    gi = 0
+    for args in cases:
+        res = dn(*args)
+        assert res.code != 0, args
+        mine = (res.out + res.err).splitlines()[:2]
+        want = golden[gi:gi + len(mine)]
+        assert mine == want, (args, mine, want)
+        gi += len(mine)
+
+    # unsupported data format: stored at add time, rejected at scan
+    assert dn("datasource-remove", "input").code == 0
+    assert dn("datasource-add", "--path=" + one,
+              "--data-format=junk", "input").code == 0
+    res = dn("scan", "input")
+    assert res.code != 0
+    assert (res.out + res.err).splitlines()[:1] == golden[gi:gi + 1]
+
+
 def test_scan_file_goldens(dn):
     """tst.scan_file.sh.out: 26 scan_testcases sections against
     one.log, then 4 sections under a datasource filter

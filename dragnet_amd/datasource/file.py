@@ -113,7 +113,8 @@ class FileDatasource(object):
     def scan(self, query, dry_run=False, out=None):
         """Run one query over the raw data.  Returns a ScanResult with a
         single aggregator (or prints the file list for dry runs)."""
-        if self.ds.data_format not in ("json", "json-skinner"):
+        from ..config import VALID_FORMATS
+        if self.ds.data_format not in VALID_FORMATS:
             # validated at use time, like the reference's parserFor
             # (lib/dragnet-impl.js:131-140; tst.badargs.sh)
             raise ValueError(

@@ -337,7 +337,12 @@ class FileDatasource(object):
         for path, _st in files:
             iq = IndexQuerier(path)
             try:
-                for mi, met in enumerate(iq.metrics):
+                for met in iq.metrics:
+                    # tag with the STORED id (== write-time position in
+                    # the metrics list), never the enumeration index:
+                    # the two must agree with the table name even if
+                    # SQLite returns dragnet_metrics rows out of
+                    # insertion order
                     tbl = "dragnet_index_%d" % met["id"]
                     cols = [b["name"] for b in met["params"]]
                     sql = "SELECT %s from %s" % (
@@ -347,7 +352,7 @@ class FileDatasource(object):
                         fields = {}
                         for c, v in zip(cols, row[:-1]):
                             fields[c] = v
-                        fields["__dn_metric"] = mi
+                        fields["__dn_metric"] = met["id"]
                         yield {"fields": fields, "value": row[-1]}
             finally:
                 iq.close()

@@ -52,6 +52,7 @@ BUCKET_LIN = 2
 
 MAX_FIELDS = 24
 MAX_DEPTH = 12
+MAX_KEY = 8  # device key tuple width (common.h / gpu.py)
 
 
 def fnv1a(data, h=FNV_OFFSET):
@@ -263,6 +264,13 @@ def compile_plan(queries, ds_filter=None, time_field=None,
     synth_req = []
 
     for q in queries:
+        if len(q.breakdowns) > MAX_KEY:
+            # the device key tuple is uint32_t key[MAX_KEY]; a wider
+            # query would write out of bounds — the CPU
+            # engine serves such queries instead
+            raise PlanError(
+                "query has more than %d breakdowns (GPU key width); "
+                "use the CPU engine" % MAX_KEY)
         prog_id = add_program(q.filter)
 
         # synthetic fields this metric needs, in the reference's

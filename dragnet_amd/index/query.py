@@ -45,7 +45,8 @@ class IndexQuerier(object):
         if not str(version).startswith("2."):
             raise IndexError_('unsupported index version: "%s"' % version)
 
-        rows = self.db.execute("SELECT * FROM dragnet_metrics").fetchall()
+        rows = self.db.execute(
+            "SELECT * FROM dragnet_metrics ORDER BY id").fetchall()
         for r in rows:
             filt = json.loads(r["filter"]) if r["filter"] is not None \
                 else None

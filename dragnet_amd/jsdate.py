@@ -86,6 +86,9 @@ def parse_ms(s):
         return None
     if hh > 24 or mm > 59 or ss > 59:
         return None
+    if hh == 24 and (mm or ss or ms):
+        # V8 accepts hour 24 only as exactly 24:00:00.000
+        return None
 
     days = days_from_civil(year, month, day)
     total = ((days * 24 + hh) * 60 + mm) * 60 + ss

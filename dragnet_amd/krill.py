@@ -87,12 +87,14 @@ def to_number(v):
         if low in ("nan", "inf", "+inf", "-inf", "infinity",
                    "+infinity", "-infinity"):
             return _NAN
-        if low.startswith("0x") or low.startswith("-0x") \
-                or low.startswith("+0x"):
+        if low.startswith("0x"):
             try:
                 return float(int(s, 16))
             except ValueError:
                 return _NAN
+        if low.startswith("-0x") or low.startswith("+0x"):
+            # JS Number() rejects signed hex ('-0x10' -> NaN)
+            return _NAN
         if low.endswith("j") or "_" in s:
             return _NAN
         try:

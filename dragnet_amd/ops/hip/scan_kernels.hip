@@ -450,9 +450,10 @@ DEV double js_to_number(BS BV, uint32_t off, uint32_t len) {
     }
     return NAN_;
   }
-  // hex
+  // hex (unsigned form only: JS Number() rejects '-0x10'/'+0x10')
   if (j - p > 2 && BV.at(off+p) == '0' &&
       (BV.at(off+p+1) == 'x' || BV.at(off+p+1) == 'X')) {
+    if (p != i) return NAN_;
     uint64_t v = 0;
     for (uint32_t k = p + 2; k < j; k++) {
       uint8_t b = BV.at(off+k);
@@ -590,6 +591,8 @@ DEV DateOut parse_iso_ms(BS BV, uint32_t off, uint32_t len) {
   long dim = DIM_TBL[month-1] + ((month == 2 && is_leap(year)) ? 1 : 0);
   if (day < 1 || day > dim) return out;
   if (hh > 24 || mm > 59 || ss > 59) return out;
+  // V8 accepts hour 24 only as exactly 24:00:00.000
+  if (hh == 24 && (mm || ss || ms)) return out;
   long days = days_from_civil(year, month, day);
   long long total = ((days * 24 + hh) * 60 + mm) * 60 + ss;
   out.ms = total * 1000 + ms - (long long)tz_off_min * 60000;

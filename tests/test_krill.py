@@ -122,3 +122,7 @@ def test_to_number():
     assert krill.to_number("Infinity") == float("inf")
     assert krill.to_number(None) == 0.0
     assert krill.to_number(True) == 1.0
+    # JS Number() rejects signed hex
+    assert isnan(krill.to_number("-0x10"))
+    assert isnan(krill.to_number("+0x10"))
+    assert isnan(krill.to_number(" -0x10 "))

@@ -108,3 +108,16 @@ def test_decode_key():
     code_undef = plan.TAG_SPECIAL << 30 | plan.SPECIAL_UNDEF
     assert plan.decode_key([code_undef], q2, strings, numbers) \
         == ("undefined",)
+
+
+def test_breakdown_width_guard():
+    # > MAX_KEY breakdowns would overflow the device key tuple
+    # (uint32_t key[8]); compile_plan must refuse
+    import pytest
+    specs = ",".join("f%d" % i for i in range(9))
+    q = query_load(breakdown_specs=specs)
+    with pytest.raises(plan.PlanError):
+        plan.compile_plan([q])
+    ok = query_load(breakdown_specs=",".join(
+        "f%d" % i for i in range(8)))
+    plan.compile_plan([ok])  # 8 wide is fine

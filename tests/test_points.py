@@ -142,6 +142,12 @@ def test_jsdate_roundtrip():
         assert jsdate.parse_ms(s) == ms, s
     assert jsdate.parse_ms("not-a-date") is None
     assert jsdate.parse_ms("2014-13-01") is None
+    # V8 accepts hour 24 only as exactly 24:00:00.000
+    assert jsdate.parse_ms("2014-05-01T24:00:00.000Z") == \
+        1398902400000 + 86400000
+    assert jsdate.parse_ms("2014-05-01T24:30:00Z") is None
+    assert jsdate.parse_ms("2014-05-01T24:00:01Z") is None
+    assert jsdate.parse_ms("2014-05-01T24:00:00.500Z") is None
     assert jsdate.to_iso(1398902745) == "2014-05-01T00:05:45.000Z"
     assert jsdate.to_iso(1398902745.6) == "2014-05-01T00:05:45.600Z"
 

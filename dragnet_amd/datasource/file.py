@@ -368,15 +368,24 @@ def find_files(roots, counters=None):
     return ff(roots, counters=counters)
 
 
-def write_index(index_path, metrics, interval, points):
+def write_index(index_path, metrics, interval, points, partition=None):
     """Route aggregated points into per-interval IndexSinks; atomic
-    rename on flush.  Returns index file paths written."""
+    rename on flush.  Returns ALL index file paths of the tree.
+
+    partition=(rank, world): distributed builds — every rank holds the
+    full merged point set (allreduce merge), so the interval buckets
+    are round-robin assigned over the SORTED bucket list and each rank
+    materializes only the files it owns (disjoint parallel writes,
+    reference reduce semantics lib/datasource-manta.js:334-350 spread
+    across ranks).  The returned list still names the whole tree.
+    """
     from ..log import get_logger
     _log = get_logger().child("datasource-file")
     _log.info("writing index", index_path=index_path,
               interval=interval, nmetrics=len(metrics))
-    written = []
     if interval == "all":
+        if partition is not None and partition[0] != 0:
+            return [os.path.join(index_path, "all")]
         sink = IndexSink(os.path.join(index_path, "all"), metrics)
         for p in points:
             sink.write_point(p)
@@ -385,20 +394,23 @@ def write_index(index_path, metrics, interval, points):
 
     prefixlen, suffix, subdir, _pat, _step = INTERVALS[interval]
     root = os.path.join(index_path, subdir)
-    sinks = {}
+    buckets = {}
     for p in points:
         dnts = p["fields"]["__dn_ts"]
         assert isinstance(dnts, (int, float)) and not math.isnan(dnts)
         datestr = jsdate.to_iso(dnts)
-        bucketname = datestr[:prefixlen]
-        if bucketname not in sinks:
-            label = bucketname.replace("T", "-")
-            start = jsdate.parse_ms(bucketname + suffix) // 1000
-            sinks[bucketname] = IndexSink(
-                os.path.join(root, label + ".sqlite"), metrics,
-                config={"dn_start": start})
-        sinks[bucketname].write_point(p)
-    for name in sorted(sinks):
-        sinks[name].flush()
-        written.append(sinks[name].filename)
+        buckets.setdefault(datestr[:prefixlen], []).append(p)
+    written = []
+    for i, name in enumerate(sorted(buckets)):
+        label = name.replace("T", "-")
+        filename = os.path.join(root, label + ".sqlite")
+        written.append(filename)
+        if partition is not None and i % partition[1] != partition[0]:
+            continue
+        start = jsdate.parse_ms(name + suffix) // 1000
+        sink = IndexSink(filename, metrics,
+                         config={"dn_start": start})
+        for p in buckets[name]:
+            sink.write_point(p)
+        sink.flush()
     return written

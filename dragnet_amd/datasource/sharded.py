@@ -14,9 +14,21 @@ other ranks return None (the CLI suppresses their output).
 
 
 from ..distributed import (dist_env, init_process_group,
-                           merge_counter_stages, merge_points_object,
+                           merge_aggregators, merge_counter_stages,
                            shard_files)
 from .file import FileDatasource, ScanResult
+
+
+def _patch_agg_stage(stages, merged_aggs):
+    """After the cross-rank merge, the Aggregator stage's noutputs is
+    the MERGED group count, not the sum of per-rank counts."""
+    out = []
+    for name, c in stages:
+        if name == "Aggregator" and merged_aggs:
+            c = dict(c)
+            c["noutputs"] = merged_aggs[0].noutputs()
+        out.append((name, c))
+    return out
 
 
 class ShardedDatasource(FileDatasource):
@@ -52,9 +64,9 @@ class ShardedDatasource(FileDatasource):
         if self.world <= 1:
             result.stages = counters.stages() + result.stages
             return result
-        merged = merge_points_object(
-            result.aggregators, [query])
-        stages = merge_counter_stages(result.stages)
+        merged = merge_aggregators(result.aggregators, [query])
+        stages = _patch_agg_stage(
+            merge_counter_stages(result.stages), merged)
         if self.rank != 0:
             return ScanResult([], [], nonroot=True)
         return ScanResult(merged, counters.stages() + stages,
@@ -86,18 +98,24 @@ class ShardedDatasource(FileDatasource):
         if self.world <= 1:
             result.stages = counters.stages() + result.stages
             return result
-        merged = merge_points_object(result.aggregators, queries)
-        stages = merge_counter_stages(result.stages)
+        merged = merge_aggregators(result.aggregators, queries)
+        stages = _patch_agg_stage(
+            merge_counter_stages(result.stages), merged)
         if self.rank != 0:
-            return ScanResult([], [], nonroot=True)
+            # every rank holds the full merge (allreduce semantics);
+            # build() overrides consume it for partitioned index writes
+            return ScanResult(merged, stages, nonroot=True)
         return ScanResult(merged, counters.stages() + stages,
                           [p for p, _ in files])
 
     def build(self, metrics, interval="day", after_ms=None,
               before_ms=None, dry_run=False):
         """Distributed build: every rank scans its shard (the map
-        phase via the overridden scan_multi, which merges to rank 0);
-        rank 0 materializes the index tree (the reduce phase)."""
+        phase); the dense RCCL merge gives EVERY rank the full
+        aggregate, so the reduce phase — materializing the index
+        tree — is partitioned by interval bucket across ranks
+        (disjoint files written in parallel; rank 0 returns the full
+        tree listing)."""
         from .file import write_index
         if not self.ds.index_path:
             raise ValueError(
@@ -108,18 +126,17 @@ class ShardedDatasource(FileDatasource):
                                  dry_run=dry_run)
         if result is None:
             return None
-        written = None
-        if not result.nonroot:
-            points = []
-            for qi, agg in enumerate(result.aggregators):
-                for p in agg.points():
-                    p["fields"]["__dn_metric"] = qi
-                    points.append(p)
-            written = write_index(self.ds.index_path, metrics,
-                                  interval, points)
+        points = []
+        for qi, agg in enumerate(result.aggregators):
+            for p in agg.points():
+                p["fields"]["__dn_metric"] = qi
+                points.append(p)
+        partition = (self.rank, self.world) if self.world > 1 else None
+        written = write_index(self.ds.index_path, metrics, interval,
+                              points, partition=partition)
         if self.dist is not None:
-            self.dist.barrier()
-        return written
+            self.dist.barrier()  # all files in place before any return
+        return None if result.nonroot else written
 
     def query(self, query, interval="day", dry_run=False, out=None):
         """Index queries run on rank 0 only (indexes are tiny relative

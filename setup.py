@@ -23,7 +23,13 @@ for stale in glob.glob("dragnet_amd/ops/hip/*_hip.hip"):
     os.unlink(stale)
 shutil.rmtree("build/temp.linux-x86_64-3.10", ignore_errors=True)
 
+import pybind11  # noqa: E402
+from setuptools import Extension  # noqa: E402
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+# The image ships libsqlite3.so.0 without a dev symlink; link the
+# versioned .so by absolute path (same image on the GPU boxes).
+_SQLITE_SO = "/usr/lib/x86_64-linux-gnu/libsqlite3.so.0"
 
 setup(
     name="dragnet_amd_ops",
@@ -38,7 +44,16 @@ setup(
                         + (["-DDN_DIRECT_BYTES"]
                            if os.environ.get("DN_DIRECT_BYTES") else []),
             },
-        )
+        ),
+        Extension(
+            name="dragnet_amd.index._csink",
+            sources=["dragnet_amd/index/csink.cpp"],
+            include_dirs=[pybind11.get_include()],
+            extra_compile_args=["-O3", "-std=c++17",
+                                "-fvisibility=hidden"],
+            extra_link_args=[_SQLITE_SO],
+            language="c++",
+        ),
     ],
     cmdclass={"build_ext": BuildExtension},
 )

@@ -50,3 +50,65 @@ def test_rebuild_clobbers(tmp_path):
     rows = list(iq.db.execute("SELECT host, value FROM dragnet_index_0"))
     iq.close()
     assert [tuple(r) for r in rows] == [("a", 2)]
+
+
+def test_native_vs_python_sink_identical(tmp_path):
+    """The native C-API sink and the pure-Python sqlite3 sink must
+    produce byte-equivalent logical content (schema + rows)."""
+    import os as _os
+    import sqlite3
+
+    from dragnet_amd.index import sink as sinkmod
+    metrics = [
+        {"name": "m", "filter": {"eq": ["a", "x"]},
+         "breakdowns": [{"name": "host", "field": "host"},
+                        {"name": "lat", "field": "lat",
+                         "aggr": "quantize"},
+                        {"name": "t", "field": "t", "date": "",
+                         "aggr": "lquantize", "step": 86400}]},
+        {"name": "m2", "filter": None,
+         "breakdowns": [{"name": "op", "field": "op"}]},
+    ]
+    points = []
+    for i in range(500):
+        points.append({"fields": {"__dn_metric": 0,
+                                  "host": "h%d" % (i % 7),
+                                  "lat": 2 ** (i % 10),
+                                  "t": 86400 * (i % 3)},
+                       "value": i + 1})
+        points.append({"fields": {"__dn_metric": 1,
+                                  "op": "op%d" % (i % 5)},
+                       "value": 2})
+
+    def build(path, force_py):
+        if force_py:
+            _os.environ["DRAGNET_PY_SINK"] = "1"
+        else:
+            _os.environ.pop("DRAGNET_PY_SINK", None)
+        try:
+            s = sinkmod.IndexSink(path, metrics,
+                                  config={"dn_start": 123})
+            assert (s._cs is None) == force_py
+            for p in points:
+                s.write_point(p)
+            s.flush()
+        finally:
+            _os.environ.pop("DRAGNET_PY_SINK", None)
+
+    f_native = str(tmp_path / "native.sqlite")
+    f_py = str(tmp_path / "py.sqlite")
+    build(f_native, False)
+    build(f_py, True)
+
+    def dump(path):
+        db = sqlite3.connect(path)
+        out = {}
+        for (tbl,) in db.execute(
+                "SELECT name FROM sqlite_master WHERE type='table' "
+                "ORDER BY name"):
+            rows = db.execute("SELECT * FROM %s" % tbl).fetchall()
+            out[tbl] = sorted(map(tuple, rows))
+        db.close()
+        return out
+
+    assert dump(f_native) == dump(f_py)

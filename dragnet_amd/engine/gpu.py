@@ -74,9 +74,12 @@ class GpuEngine(object):
         dict_slots = _env_int("DRAGNET_DICT_SLOTS", 1 << 20)
         dict_data = _env_int("DRAGNET_DICT_DATA_MB", 256) * 1024 * 1024
 
-        for attempt in range(3):
+        # byte sources cannot restart, so they never risk the dense
+        # directory (its capacity bounds key cardinality)
+        dense = False if byte_source is not None else None
+        for attempt in range(4):
             ctx = _ScanContext(self, cplan, agg_slots, dict_slots,
-                               dict_data)
+                               dict_data, dense=dense)
             if byte_source is not None:
                 if attempt > 0:
                     raise RuntimeError(
@@ -88,9 +91,18 @@ class GpuEngine(object):
                 ctx.scan_files(files)
             if ctx.overflowed():
                 from ..log import get_logger
+                if ctx.dense:
+                    # dense directory overflowed (high cardinality):
+                    # restart on the atomic hash path at full size
+                    get_logger().child("gpu-engine").warn(
+                        "dense directory overflow; restarting on the "
+                        "hash path", agg_slots=agg_slots)
+                    dense = False
+                    continue
                 get_logger().child("gpu-engine").warn(
                     "aggregation table overflow; restarting scan",
                     agg_slots=agg_slots * 8, dict_slots=dict_slots * 8)
+                dense = False
                 agg_slots *= 8
                 dict_slots *= 8
                 dict_data *= 4
@@ -100,13 +112,28 @@ class GpuEngine(object):
         raise RuntimeError("aggregation tables overflowed after retries")
 
 
+PROWS = 2048  # dense partial rows == the scan kernel's grid cap
+
+
 class _ScanContext(object):
-    def __init__(self, eng, cplan, agg_slots, dict_slots, dict_data_cap):
+    def __init__(self, eng, cplan, agg_slots, dict_slots, dict_data_cap,
+                 dense=None):
         torch = eng.torch
         self.eng = eng
         self.t = torch
         self.cplan = cplan
         dev = eng.device
+        # Dense-accumulation mode (default on): the aggregate table is
+        # a small slot directory, per-workgroup counts go to a dense
+        # [PROWS, slots] partial matrix, and the MFMA f64 column-sum
+        # reduce folds it before extraction.  High
+        # cardinality overflows the directory and the engine restarts
+        # with dense off (the atomic hash path).
+        if dense is None:
+            dense = _env_int("DRAGNET_DENSE", 1) == 1
+        self.dense = dense
+        if dense:
+            agg_slots = _env_int("DRAGNET_DENSE_SLOTS", 4096)
         i32 = dict(dtype=torch.int32, device=dev)
         f64 = dict(dtype=torch.float64, device=dev)
         u8 = dict(dtype=torch.uint8, device=dev)
@@ -141,14 +168,18 @@ class _ScanContext(object):
 
         # tables
         self.tables = []
+        self.partials = []
         for m in range(self.nm):
             state = torch.zeros(agg_slots, **i32)
             keys = torch.zeros(agg_slots * MAX_KEY, **i32)
             count = torch.zeros(agg_slots, **f64)
             self.tables.append((state, keys, count))
+            if dense:
+                self.partials.append(
+                    torch.zeros((PROWS, agg_slots), **f64))
         descs = eng.ops.agg_descs_host(
             [t[0] for t in self.tables], [t[1] for t in self.tables],
-            [t[2] for t in self.tables])
+            [t[2] for t in self.tables], self.partials)
         self.table_descs = descs.to(dev)
 
         # dictionaries
@@ -597,7 +628,8 @@ class _ScanContext(object):
         self.eng.ops.scan_reset(
             [t[0] for t in self.tables], [t[2] for t in self.tables],
             self.sd["state"], self.sd["used"], self.sd["next"],
-            self.nd["state"], self.nd["next"], self.counters)
+            self.nd["state"], self.nd["next"], self.counters,
+            self.partials)
 
     # ---- results ----
 
@@ -620,7 +652,8 @@ class _ScanContext(object):
             self.sd["off"], self.sd["len"], self.sd["data"],
             self.sd["used"], self.sd["next"],
             self.nd["state"], self.nd["bits"], self.nd["id"],
-            self.nd["next"], self.counters, self.dict_slots)
+            self.nd["next"], self.counters, self.dict_slots,
+            self.partials)
         h = {"cnt": r[0], "n_str": r[1], "n_num": r[2], "used": r[3],
              "blob": r[4], "str_off": r[5], "str_len": r[6],
              "numbers": r[7],

@@ -75,11 +75,22 @@ constexpr uint32_t SLOT_EMPTY = 0, SLOT_CLAIMED = 1, SLOT_READY = 2;
 //   state: u32[nslots]
 //   keys:  u32[nslots][MAX_KEY]
 //   count: double[nslots]
+//
+// Dense-accumulation mode (partial != nullptr): the table is only a
+// slot DIRECTORY (insert assigns a stable slot, no count atomics);
+// each workgroup accumulates into its own row of the dense
+// partial[prows][nslots] matrix and the MFMA column-sum reduce
+// (mfma_reduce_kernel, v_mfma_f64_16x16x4_f64) folds the matrix into
+// count[] before extraction — the matrix-shaped accumulation of
+// SURVEY §7.7 / the BASELINE north star.
 struct AggTable {
   uint32_t* state;
   uint32_t* keys;   // nslots * MAX_KEY
   double*   count;
   uint32_t  nslots; // power of two
+  double*   partial;  // [prows][nslots] per-workgroup rows (or null)
+  uint32_t  prows;
+  uint32_t  pad_;
 };
 
 // String-intern table (shared by all metrics):

@@ -54,7 +54,26 @@ AggTable make_table(const torch::Tensor& state, const torch::Tensor& keys,
   t.keys = (uint32_t*)keys.data_ptr();
   t.count = (double*)count.data_ptr();
   t.nslots = (uint32_t)state.numel();
+  t.partial = nullptr;
+  t.prows = 0;
+  t.pad_ = 0;
   return t;
+}
+
+// Fold a dense partial matrix into count[] with the MFMA column-sum
+// reduce (must run before extraction of a dense-mode table).
+void dense_reduce_one(const AggTable& T, hipStream_t stream) {
+  if (T.partial == nullptr || T.prows == 0) return;
+  uint32_t tiles = (T.nslots + 15) / 16;
+  uint32_t splits = 32;
+  while (splits > 1 && T.prows / splits < 16) splits /= 2;
+  uint32_t rows_per_blk = (T.prows + splits - 1) / splits;
+  rows_per_blk = (rows_per_blk + 3) & ~3u;  // multiple of the K=4 step
+  hipLaunchKernelGGL(mfma_reduce_kernel, dim3(tiles, splits), dim3(64),
+                     0, stream, T, rows_per_blk);
+  hipError_t err = hipGetLastError();
+  TORCH_CHECK(err == hipSuccess, "mfma_reduce launch failed: ",
+              hipGetErrorString(err));
 }
 
 hipStream_t current_stream() {
@@ -253,7 +272,8 @@ void scan_reset(std::vector<torch::Tensor> states,
                 std::vector<torch::Tensor> counts,
                 torch::Tensor sd_state, torch::Tensor sd_used,
                 torch::Tensor sd_next, torch::Tensor nd_state,
-                torch::Tensor nd_next, torch::Tensor counters) {
+                torch::Tensor nd_next, torch::Tensor counters,
+                std::vector<torch::Tensor> partials) {
   hipStream_t st = current_stream();
   auto z = [&](torch::Tensor& t) {
     hipError_t e = hipMemsetAsync(t.data_ptr(), 0,
@@ -264,21 +284,32 @@ void scan_reset(std::vector<torch::Tensor> states,
   };
   for (auto& t : states) z(t);
   for (auto& t : counts) z(t);
+  for (auto& t : partials) z(t);
   z(sd_state); z(sd_used); z(sd_next);
   z(nd_state); z(nd_next); z(counters);
 }
 
 // Build the device-side AggTable descriptor array from per-metric
 // tensors.  Returns a CPU byte tensor; the caller copies it to the GPU.
+// partials (optional, same length): [prows, nslots] f64 dense partial
+// matrices — enables the dense-accumulation + MFMA-reduce path.
 torch::Tensor agg_descs_host(std::vector<torch::Tensor> states,
                              std::vector<torch::Tensor> keys,
-                             std::vector<torch::Tensor> counts) {
+                             std::vector<torch::Tensor> counts,
+                             std::vector<torch::Tensor> partials) {
   int nm = (int)states.size();
   auto out = torch::empty({(long)(nm * sizeof(AggTable))},
                           torch::dtype(torch::kUInt8));
   AggTable* descs = (AggTable*)out.data_ptr();
-  for (int m = 0; m < nm; m++)
+  for (int m = 0; m < nm; m++) {
     descs[m] = make_table(states[m], keys[m], counts[m]);
+    if (!partials.empty()) {
+      TORCH_CHECK(partials[m].size(1) == (long)descs[m].nslots,
+                  "partial width != nslots");
+      descs[m].partial = (double*)partials[m].data_ptr();
+      descs[m].prows = (uint32_t)partials[m].size(0);
+    }
+  }
   return out;
 }
 
@@ -386,7 +417,16 @@ std::vector<torch::Tensor> extract_all(
     torch::Tensor sd_off, torch::Tensor sd_len, torch::Tensor sd_data,
     torch::Tensor sd_used, torch::Tensor sd_next,
     torch::Tensor nd_state, torch::Tensor nd_bits, torch::Tensor nd_id,
-    torch::Tensor nd_next, torch::Tensor counters, int64_t dict_slots) {
+    torch::Tensor nd_next, torch::Tensor counters, int64_t dict_slots,
+    std::vector<torch::Tensor> partials) {
+  // dense mode: fold the per-workgroup partial matrices into count[]
+  // (MFMA column-sum) before the table extraction below reads them
+  for (size_t m = 0; m < partials.size(); m++) {
+    AggTable T = make_table(states[m], keys[m], counts[m]);
+    T.partial = (double*)partials[m].data_ptr();
+    T.prows = (uint32_t)partials[m].size(0);
+    dense_reduce_one(T, current_stream());
+  }
   std::vector<torch::Tensor> out;
   out.push_back(counters.clone());
   out.push_back(sd_next.clone());

@@ -1192,6 +1192,98 @@ DEV bool agg_insert(const AggTable& T, const uint32_t* key, int nk,
   return false;  // pathological probe chain: table too full
 }
 
+// Directory variant for the dense-accumulation path: assign (or find)
+// the key's stable slot WITHOUT touching count — counts accumulate in
+// the caller's per-workgroup dense partial row and are folded by the
+// MFMA reduce.  Returns the slot index or ~0u on probe exhaustion.
+DEV uint32_t agg_insert_slot(const AggTable& T, const uint32_t* key,
+                             int nk) {
+  uint64_t h = FNV_OFFSET;
+  for (int k = 0; k < MAX_KEY; k++)
+    h = mix64(h ^ (k < nk ? key[k] : 0u) ^ (uint64_t)(k + 1) * 0x9E3779B97F4A7C15ull);
+  uint32_t mask = T.nslots - 1;
+  uint32_t s = (uint32_t)h & mask;
+  // probe cap well under nslots: a directory near 100% load degrades
+  // to O(n) probes — declare overflow early and let the engine restart
+  // on the atomic hash path instead
+  uint32_t max_probes = T.nslots / 4 + 64;
+  for (uint32_t probes = 0; probes < max_probes; probes++, s = (s + 1) & mask) {
+    while (true) {
+      uint32_t st = atomic_load_relaxed(&T.state[s]);
+      if (st == SLOT_READY) {
+        bool same = true;
+#pragma unroll
+        for (int k = 0; k < MAX_KEY; k++)
+          if (k < nk &&
+              atomic_load_relaxed(&T.keys[s * MAX_KEY + k]) != key[k])
+            same = false;
+        if (same) return s;
+        break;
+      }
+      if (st == SLOT_EMPTY) {
+        uint32_t prev = atomicCAS(&T.state[s], SLOT_EMPTY, SLOT_CLAIMED);
+        if (prev == SLOT_EMPTY) {
+#pragma unroll
+          for (int k = 0; k < MAX_KEY; k++)
+            atomic_store_relaxed(&T.keys[s * MAX_KEY + k],
+                                 k < nk ? key[k] : 0u);
+          drain_stores();
+          atomic_store_relaxed(&T.state[s], SLOT_READY);
+          return s;
+        }
+        continue;
+      }
+      __builtin_amdgcn_s_sleep(1);
+    }
+  }
+  return 0xFFFFFFFFu;
+}
+
+// Add one (key, weight) into a table: dense path (directory slot +
+// this workgroup's partial row) or the atomic hash path.
+DEV bool agg_add(const AggTable& T, const uint32_t* key, int nk,
+                 double w) {
+  if (T.partial != nullptr) {
+    uint32_t s = agg_insert_slot(T, key, nk);
+    if (s == 0xFFFFFFFFu) return false;
+    atomicAdd(&T.partial[(size_t)(blockIdx.x % T.prows) * T.nslots + s],
+              w);
+    return true;
+  }
+  return agg_insert(T, key, nk, w);
+}
+
+// MFMA column-sum reduce: count[s] += sum_r partial[r][s], computed as
+// ones[16,4] x partial-tile[4,16] on the f64 matrix core
+// (v_mfma_f64_16x16x4_f64; A = all-ones so every accumulator row holds
+// the column sum).  One wave per 16-slot tile x gridDim.y row splits.
+typedef double dn_d4 __attribute__((ext_vector_type(4)));
+
+__global__ void mfma_reduce_kernel(AggTable T, uint32_t rows_per_blk) {
+  uint32_t slot0 = blockIdx.x * 16;
+  uint32_t r0 = blockIdx.y * rows_per_blk;
+  uint32_t r1 = r0 + rows_per_blk;
+  if (r1 > T.prows) r1 = T.prows;
+  int l = threadIdx.x;
+  uint32_t n = slot0 + (l & 15);
+  dn_d4 acc = {0.0, 0.0, 0.0, 0.0};
+  const double* P = T.partial;
+  size_t stride = T.nslots;
+  for (uint32_t rt = r0; rt < r1; rt += 4) {
+    uint32_t rr = rt + (uint32_t)(l >> 4);
+    // B fragment: lane l holds B[k = l>>4][j = l&15]
+    double b = (rr < r1 && n < T.nslots)
+                   ? P[(size_t)rr * stride + n] : 0.0;
+    acc = __builtin_amdgcn_mfma_f64_16x16x4f64(1.0, b, acc, 0, 0, 0);
+  }
+  // C/D map: lane l holds D[(l>>4)*4 + reg, l&15]; rows identical
+  // (A = ones), so lanes 0..15 publish their column's sum
+  if (l < 16 && n < T.nslots) {
+    double v = acc[0];
+    if (v != 0.0) atomicAdd(&T.count[n], v);
+  }
+}
+
 // -------------------------------------------------------------------
 // the fused scan kernel
 
@@ -1602,7 +1694,7 @@ DEV void process_record(BS BV, uint32_t start, uint32_t end,
           ci = (ci + 1) & (LDS_CACHE - 1);
         }
         if (!cached) {
-          if (!agg_insert(A.tables[m], key, nk, weight))
+          if (!agg_add(A.tables[m], key, nk, weight))
             atomicAdd(&lcnt[C_OVERFLOW], 1ull);
         }
       }
@@ -1722,9 +1814,9 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
   __syncthreads();
   for (int i = threadIdx.x; i < LDS_CACHE; i += BLOCK) {
     if (cache[i].hash != 0) {
-      if (!agg_insert(A.tables[cache[i].metric], cache[i].key,
-                      P.metric_rows[cache[i].metric * 8 + 1],
-                      cache[i].count))
+      if (!agg_add(A.tables[cache[i].metric], cache[i].key,
+                   P.metric_rows[cache[i].metric * 8 + 1],
+                   cache[i].count))
         atomicAdd(&lcnt[C_OVERFLOW], 1ull);
     }
   }

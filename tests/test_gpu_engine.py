@@ -457,3 +457,57 @@ def test_many_chunk_boundaries(engines, tmp_path):
     assert gs["ninputs"] == 20_000
     assert gs["invalid json"] == 0
     assert g.aggregators[0].points() == c.aggregators[0].points()
+
+
+def test_dense_mfma_vs_hash_path(engines, tmp_path):
+    """The dense-accumulation path (slot directory + per-workgroup
+    partial matrix + MFMA f64 column-sum reduce) must equal the atomic
+    hash path and the CPU oracle exactly — including near the dense
+    directory's capacity and with json-skinner weights."""
+    import json as _json
+    import random
+
+    from dragnet_amd.engine.gpu import GpuEngine
+    from dragnet_amd.query import query_load
+    cpu, _ = engines
+
+    rng = random.Random(7)
+    lines = []
+    # ~3000 distinct (a, b) keys: well into the directory, near the
+    # 4096-slot capacity, all slots exercised by the MFMA reduce
+    for i in range(120000):
+        rec = {"a": "k%d" % rng.randint(0, 120),
+               "b": rng.randint(0, 24),
+               "lat": rng.choice([3, 17, "99", 2 ** rng.randint(0, 20)])}
+        lines.append(_json.dumps(rec))
+    f = tmp_path / "dense.log"
+    f.write_bytes(("\n".join(lines) + "\n").encode())
+
+    q = query_load(breakdown_specs="a,b,lat[aggr=quantize]")
+    c = cpu.scan([str(f)], [q])
+    os.environ["DRAGNET_DENSE"] = "1"
+    try:
+        g_dense = GpuEngine().scan([str(f)], [q])
+    finally:
+        os.environ["DRAGNET_DENSE"] = "0"
+    try:
+        g_hash = GpuEngine().scan([str(f)], [q])
+    finally:
+        os.environ.pop("DRAGNET_DENSE", None)
+    assert_same(c, g_dense)
+    assert_same(c, g_hash)
+
+
+def test_dense_overflow_falls_back(engines, tmp_path):
+    """Cardinality beyond the dense directory restarts on the hash
+    path and still matches the oracle."""
+    import json as _json
+
+    cpu, gpu = engines
+    lines = [_json.dumps({"u": "id%06d" % i}) for i in range(30000)]
+    f = tmp_path / "hi_card.log"
+    f.write_bytes(("\n".join(lines) + "\n").encode())
+    q = queries_from_case(["-b", "u"])
+    c = cpu.scan([str(f)], [q])
+    g = gpu.scan([str(f)], [q])
+    assert_same(c, g)

@@ -32,9 +32,10 @@ def parse_json_line(line):
 
 
 class FilterStage(object):
-    def __init__(self, name, predicate):
+    def __init__(self, name, predicate, warn=None):
         self.name = name
         self.pred = predicate
+        self.warn = warn
         self.counters = {"ninputs": 0, "noutputs": 0,
                          "nfilteredout": 0, "nfailedeval": 0}
 
@@ -43,11 +44,12 @@ class FilterStage(object):
         c["ninputs"] += 1
         try:
             ok = self.pred.eval(fields)
-        except krill.MissingFieldError:
+        except Exception as e:
+            # vsWarn('nfailedeval') — reference
+            # lib/krill-skinner-stream.js:43-45
             c["nfailedeval"] += 1
-            return False
-        except Exception:
-            c["nfailedeval"] += 1
+            if self.warn is not None:
+                self.warn(str(e), self.name, c["ninputs"])
             return False
         if not ok:
             c["nfilteredout"] += 1
@@ -62,11 +64,16 @@ class SyntheticStage(object):
     through, strings Date.parse -> floor(ms/1000); first failure per
     record is counted ('undef' or 'baddate') and the record dropped."""
 
-    def __init__(self, synthetic):
+    def __init__(self, synthetic, warn=None):
         self.name = "Datetime parser"
         self.synthetic = synthetic
+        self.warn = warn
         self.counters = {"ninputs": 0, "noutputs": 0,
                          "undef": 0, "baddate": 0}
+
+    def _warn(self, msg):
+        if self.warn is not None:
+            self.warn(msg, self.name, self.counters["ninputs"])
 
     def accept(self, fields):
         c = self.counters
@@ -77,12 +84,16 @@ class SyntheticStage(object):
             if val is krill.MISSING:
                 if nerrors == 0:
                     c["undef"] += 1
+                    # reference lib/stream-synthetic.js:50-52
+                    self._warn('field "%s" is undefined' % fc["field"])
                 nerrors += 1
                 continue
             if isinstance(val, bool):
                 # typeof bool != 'number' in JS -> Date.parse(bool) -> NaN
                 if nerrors == 0:
                     c["baddate"] += 1
+                    self._warn('field "%s" is not a valid date'
+                               % fc["field"])
                 nerrors += 1
                 continue
             if isinstance(val, (int, float)):
@@ -92,6 +103,9 @@ class SyntheticStage(object):
             if ms is None:
                 if nerrors == 0:
                     c["baddate"] += 1
+                    # reference lib/stream-synthetic.js:70-72
+                    self._warn('field "%s" is not a valid date'
+                               % fc["field"])
                 nerrors += 1
                 continue
             fields[fc["name"]] = ms // 1000
@@ -111,21 +125,31 @@ class ScanPipeline(object):
         data_format  'json' | 'json-skinner'
     """
 
+    WARN_CAP = 1000  # per-record warning entries retained
+
     def __init__(self, query, ds_filter=None, time_field=None,
-                 data_format="json"):
+                 data_format="json", collect_warnings=False):
         self.query = query
         self.data_format = data_format
         self.parser_counters = {"ninputs": 0, "noutputs": 0,
                                 "invalid json": 0}
         self.stages = []
+        # per-record vstream-style warnings: (message, context label)
+        # pairs, emitted by each stage as records drop (reference
+        # bin/dn warn(): 'warn: <msg>\n    at <context.label()>')
+        self.warnings = []
+        self.context_file = None  # set by the engine per input file
+        warn = self._warn_record if collect_warnings else None
 
         if ds_filter is not None:
             self.stages.append(FilterStage(
-                "Datasource filter", krill.create_predicate(ds_filter)))
+                "Datasource filter", krill.create_predicate(ds_filter),
+                warn=warn))
 
         if query.filter is not None:
             self.stages.append(FilterStage(
-                "User filter", krill.create_predicate(query.filter)))
+                "User filter", krill.create_predicate(query.filter),
+                warn=warn))
 
         synthetic = list(query.synthetic)
         if query.before_ms is not None or query.after_ms is not None:

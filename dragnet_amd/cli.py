@@ -12,6 +12,11 @@ import json
 import os
 
 import sys
+import time as _time
+
+# "require phase" timing (-t; reference bin/dn:8,24,80-83): time spent
+# importing the engine modules below, reported separately from total
+_t_require0 = _time.monotonic()
 
 from . import config as mod_config
 from . import krill
@@ -19,6 +24,8 @@ from . import output as mod_output
 from .attrs import AttrsError, attrs_parse
 from .datasource import datasource_for_config
 from .query import QueryConfig, QueryError
+
+_t_require = _time.monotonic() - _t_require0
 
 USAGE = """\
 usage: dn SUBCOMMAND [OPTIONS] ARGS
@@ -88,6 +95,7 @@ def main(argv=None):
     if track_time:
         total = time.monotonic() - t_start
         sys.stderr.write("timing stats:\n")
+        sys.stderr.write("    require:  %.6fs\n" % _t_require)
         sys.stderr.write("    total:    %.6fs\n" % total)
     return rv
 
@@ -491,9 +499,41 @@ def _query_from_opts(opts, allow_reserved=False):
     return qc
 
 
+# record-pipeline stages whose counters must conserve records
+# (aggregators emit distinct groups, find stages expand directories)
+_CONSERVING_STAGES = frozenset([
+    "json parser", "SkinnerAdapterStream", "Datasource filter",
+    "User filter", "Datetime parser", "Time filter"])
+_DROP_KINDS = ("invalid json", "nfilteredout", "nfailedeval", "undef",
+               "baddate", "nonnumeric")
+
+
+def _integrity_guard(result):
+    """Premature-exit guard analog (reference bin/dn:1276-1311): the
+    reference fails a run that exits 0 without completing its pipeline
+    (dropped callbacks).  The synchronous analog of 'work silently
+    vanished' is a conservation violation — a stage whose ninputs !=
+    noutputs + attributed drops.  On violation: the reference's error
+    wording, a full counter dump, exit 1."""
+    for name, c in result.stages:
+        if name not in _CONSERVING_STAGES:
+            continue
+        if "ninputs" not in c or "noutputs" not in c:
+            continue
+        drops = sum(c.get(k, 0) for k in _DROP_KINDS)
+        if c["ninputs"] != c["noutputs"] + drops:
+            sys.stderr.write("ERROR: internal error: premature exit\n")
+            mod_output.dump_counters(result.stages, out=sys.stderr)
+            raise FatalError(
+                'stage "%s" dropped records without attribution '
+                "(ninputs=%d noutputs=%d drops=%d)"
+                % (name, c["ninputs"], c["noutputs"], drops))
+
+
 def _output_result(query, opts, result, title=None):
     """Render a ScanResult per the raw/points/gnuplot/pretty options
     (reference dnOutput, bin/dn:924-967)."""
+    _integrity_guard(result)
     agg = result.aggregators[0]
     npoints = agg.noutputs()
     if opts.get("points"):
@@ -517,17 +557,26 @@ def _output_result(query, opts, result, title=None):
         mod_output.dump_counters(stages)
 
     if opts.get("warnings"):
-        # summary of warn-and-drop events (per-stage drop counters)
-        # (plain filter-outs are not warnings; reference vsWarn kinds)
-        drop_kinds = ("invalid json", "nfailedeval", "undef",
-                      "baddate", "nonnumeric")
-        for name, counters in result.stages:
-            for kind in drop_kinds:
-                v = counters.get(kind, 0)
-                if v:
-                    sys.stderr.write(
-                        "warn: %s: %s: %d record%s dropped\n"
-                        % (name, kind, v, "s" if v != 1 else ""))
+        # per-record vstream-style warnings with context chains
+        # (reference bin/dn warn(): 'warn: <msg>\n    at <label>',
+        # bin/dn:135-144); engines that only count drops (the GPU
+        # kernel) fall back to per-stage counter summaries in the
+        # same shape
+        printed = False
+        for msg, label in getattr(result, "warnings", []) or []:
+            sys.stderr.write("warn: %s\n    at %s\n" % (msg, label))
+            printed = True
+        if not printed:
+            drop_kinds = ("invalid json", "nfailedeval", "undef",
+                          "baddate", "nonnumeric")
+            for name, counters in result.stages:
+                for kind in drop_kinds:
+                    v = counters.get(kind, 0)
+                    if v:
+                        sys.stderr.write(
+                            "warn: %d record%s dropped (%s)\n"
+                            "    at %s\n"
+                            % (v, "s" if v != 1 else "", kind, name))
 
 
 def cmd_scan(argv):
@@ -535,6 +584,9 @@ def cmd_scan(argv):
         "before", "after", "filter", "breakdowns", "raw", "points",
         "counters", "warnings", "gnuplot", "dry-run"])
     check_arg_count(args, 1)
+    if opts.get("warnings"):
+        # engines check this to collect per-record warning context
+        os.environ["DRAGNET_WARNINGS"] = "1"
     cfg = _load_config()
     ds = _get_datasource(cfg, args[0])
     backend = datasource_for_config(ds)

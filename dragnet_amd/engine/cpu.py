@@ -16,11 +16,15 @@ class CpuEngine(object):
         """Run `queries` over the concatenated bytes of `files` (or over
         `byte_source`, an iterable of byte chunks).  Returns a
         ScanResult-compatible object."""
+        import os
+
         from ..datasource.file import ScanResult
 
+        collect = os.environ.get("DRAGNET_WARNINGS") == "1"
         pipelines = [
             ScanPipeline(q, ds_filter=ds_filter, time_field=time_field,
-                         data_format=data_format)
+                         data_format=data_format,
+                         collect_warnings=collect)
             for q in queries
         ]
         primary = pipelines[0]
@@ -43,7 +47,7 @@ class CpuEngine(object):
             return consumed
 
         if byte_source is None:
-            byte_source = _read_files(files)
+            byte_source = _read_files(files, pipelines=pipelines)
 
         partial = b""
         for chunk in byte_source:
@@ -56,11 +60,16 @@ class CpuEngine(object):
             feed_line(partial)
 
         stages = pipelines[0].counter_stages()
-        return ScanResult([p.aggr for p in pipelines], stages)
+        result = ScanResult([p.aggr for p in pipelines], stages)
+        result.warnings = primary.warnings
+        return result
 
 
-def _read_files(files, chunk_size=8 * 1024 * 1024):
+def _read_files(files, chunk_size=8 * 1024 * 1024, pipelines=None):
     for path in files:
+        # warning context: which file the following records came from
+        for p in pipelines or []:
+            p.context_file = path
         with open(path, "rb", buffering=0) as f:
             while True:
                 chunk = f.read(chunk_size)

@@ -160,7 +160,7 @@ class ScanPipeline(object):
             synthetic.append(
                 {"name": "dn_ts", "field": time_field, "date": ""})
         if synthetic:
-            self.stages.append(SyntheticStage(synthetic))
+            self.stages.append(SyntheticStage(synthetic, warn=warn))
 
         tbf = query.time_bounds_filter("dn_ts")
         if tbf is not None:
@@ -168,8 +168,18 @@ class ScanPipeline(object):
                 "Time filter", krill.create_predicate(tbf)))
 
         self.aggr = Aggregator(query)
+        self._collect = collect_warnings
         self._partial = b""
         self.last_point = None  # most recent successfully parsed point
+
+    def _warn_record(self, message, stage, n):
+        """Record one vstream-style warning with its context chain."""
+        if len(self.warnings) >= self.WARN_CAP:
+            return
+        label = "%s input %d" % (stage, n)
+        if self.context_file:
+            label += " (%s)" % self.context_file
+        self.warnings.append((message, label))
 
     # ---- byte-stream input ----
 
@@ -196,8 +206,12 @@ class ScanPipeline(object):
         self.parser_counters["ninputs"] += 1
         try:
             obj = parse_json_line(line)
-        except ValueError:
+        except ValueError as e:
             self.parser_counters["invalid json"] += 1
+            if self._collect:
+                self._warn_record("invalid json: %s" % e,
+                                  "json parser",
+                                  self.parser_counters["ninputs"])
             return
         self.parser_counters["noutputs"] += 1
         if self.data_format == "json-skinner":

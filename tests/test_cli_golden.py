@@ -282,3 +282,67 @@ def test_warnings_flag(dn, fixture_tree):
     r = dn("scan", "--warnings", "tree")
     assert r.code == 0
     assert "invalid json" in r.err
+
+
+def test_warnings_context_chains(dn, fixture_tree):
+    """--warnings prints per-record vstream-style warnings with
+    context chains on the CPU engine (reference bin/dn:135-144
+    'warn: <msg>\\n    at <label>')."""
+    r = dn("datasource-add", "warnsrc", "--path=" + fixture_tree,
+           "--time-field=time")
+    assert r.code == 0, r.err
+    r = dn("scan", "--warnings", "-b", "time[date,aggr=lquantize,"
+           "step=86400]", "warnsrc")
+    assert r.code == 0, r.err
+    # fixture has 2 invalid JSON lines, 1 bad date, 1 missing time
+    assert "warn: invalid json:" in r.err
+    assert 'warn: field "time" is not a valid date' in r.err
+    assert 'warn: field "time" is undefined' in r.err
+    # every warning carries a context label line
+    warns = [ln for ln in r.err.splitlines()
+             if ln.startswith("warn:")]
+    ats = [ln for ln in r.err.splitlines()
+           if ln.startswith("    at ")]
+    assert len(warns) == len(ats) >= 4
+    assert any("json parser input" in a for a in ats)
+    assert any("Datetime parser input" in a for a in ats)
+
+
+def test_timing_flag_require_split(dn, fixture_tree, capsys):
+    """-t prints require-phase and total timings (reference
+    bin/dn:80-83, 1291-1296)."""
+    from dragnet_amd import cli
+    import io
+    import sys as _sys
+    err = io.StringIO()
+    old = _sys.stderr
+    _sys.stderr = err
+    try:
+        rv = cli.main(["-t", "datasource-list"])
+    finally:
+        _sys.stderr = old
+    assert rv == 0 or rv is None
+    out = err.getvalue()
+    assert "timing stats:" in out
+    assert "require:" in out
+    assert "total:" in out
+
+
+def test_integrity_guard_fires_on_vanished_records(dn, fixture_tree):
+    """The premature-exit guard detects a stage that loses records
+    without attributing a drop counter."""
+    from dragnet_amd import cli as mod_cli
+
+    class FakeResult(object):
+        aggregators = []
+        stages = [("json parser",
+                   {"ninputs": 10, "noutputs": 8, "invalid json": 1})]
+        warnings = []
+    import pytest as _pytest
+    with _pytest.raises(mod_cli.FatalError):
+        mod_cli._integrity_guard(FakeResult())
+    # conserving counters pass
+    ok = FakeResult()
+    ok.stages = [("json parser",
+                  {"ninputs": 10, "noutputs": 9, "invalid json": 1})]
+    mod_cli._integrity_guard(ok)

@@ -100,7 +100,8 @@ void newline_index(torch::Tensor data, int64_t start, int64_t n,
   auto stream = current_stream();
   const uint8_t* d = (const uint8_t*)data.data_ptr();
   uint32_t* segs = (uint32_t*)seg_scratch.data_ptr();
-  uint32_t blocks = (nseg + 255) / 256;
+  // wave-per-segment kernels: 256-thread blocks cover 4 segments each
+  uint32_t blocks = (nseg + 3) / 4;
   hipLaunchKernelGGL(newline_count_kernel, dim3(blocks), dim3(256), 0,
                      stream, d, (uint32_t)start, (uint32_t)n, segs, nseg);
   hipLaunchKernelGGL(newline_scan_kernel, dim3(1), dim3(1024), 0,

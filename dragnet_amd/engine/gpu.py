@@ -565,16 +565,71 @@ class _ScanContext(object):
 
     # ---- wave-transposed staging (prototype; DRAGNET_XPOSE) ----
 
-    def stage_xpose(self, buf):
-        """Re-lay a byte pool wave-transposed for scan_kernel_x:
-        records length-sorted (aggregation is order-independent), 64
-        consecutive sorted records form a wave, and each record's
-        bytes are split into 64B granules interleaved so granule g of
-        lane l sits at wave_base + g*4096 + l*64 — a wave's window
-        refills then touch 64 CONSECUTIVE granules (coalesced) instead
-        of 64 scattered records (the SQ_WAIT-bound gather profile in
-        profiles/).  Prototype uses one global granule count (max
-        record length), fine for near-uniform record sizes."""
+    def stage_xpose(self, buf=None):
+        """Wave-transposed staging for scan_kernel_x: records
+        length-sorted (aggregation is order-independent), 64
+        consecutive sorted records form a wave, each record's bytes
+        split into granules interleaved so granule g of lane l sits at
+        wave_base + g*64*gran + l*gran — a wave's window refills touch
+        64 CONSECUTIVE granules (coalesced) instead of 64 scattered
+        records.
+
+        Default: built ON DEVICE from the resident pool (torch sort +
+        cumsum + the xpose_build_kernel scatter) — no host staging
+        pass.  DRAGNET_XPOSE_HOST=1 selects the r1 numpy builder
+        (A/B + layout unit tests)."""
+        if os.environ.get("DRAGNET_XPOSE_HOST") == "1":
+            return self._stage_xpose_host(buf)
+        torch = self.t
+        dev = self.eng.device
+        gran = _env_int("DRAGNET_XGRAN", 32)
+        glog = gran.bit_length() - 1
+        if not hasattr(self, "_resident"):
+            if buf is None:
+                raise ValueError("no resident pool to transpose")
+            self.stage_resident(buf)
+        n, _padded = self._resident
+        # full-pool line index (the staged slice indexes cover slices)
+        self.eng.ops.newline_index(self._dev_data, 0, n, self._segs,
+                                   self._pos, self._nlines)
+        nrec = int(self._nlines.item())  # one-time staging sync
+        if nrec == 0:
+            raise ValueError("no records")
+        pos = self._pos[:nrec].to(torch.int64)
+        starts = torch.empty_like(pos)
+        starts[0] = 0
+        starts[1:] = pos[:-1] + 1
+        lens_sorted, order = torch.sort(pos - starts)
+        starts_sorted = starts[order]
+        nslots = (nrec + 63) & ~63
+        nw = nslots // 64
+        slot_len = torch.full((nslots,), -1, dtype=torch.int32,
+                              device=dev)
+        slot_len[:nrec] = lens_sorted.to(torch.int32)
+        sstart = torch.zeros(nslots, dtype=torch.int32, device=dev)
+        sstart[:nrec] = starts_sorted.to(torch.int32)
+        # per-wave granule count from each wave's longest (last) lane
+        last = torch.clamp(
+            torch.arange(nw, device=dev, dtype=torch.int64) * 64 + 63,
+            max=nrec - 1)
+        gwl = torch.clamp((lens_sorted[last] + gran - 1) // gran, min=1)
+        stride = 64 * gran
+        wbase = torch.zeros(nw + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(gwl * stride, 0, out=wbase[1:])
+        total = int(wbase[-1].item()) + stride  # +slack
+        xb = torch.empty(total, dtype=torch.uint8, device=dev)
+        self.eng.ops.xpose_build(self._dev_data, sstart, slot_len,
+                                 wbase, nslots, glog, xb)
+        self._x = {
+            "xdata": xb,
+            "wave_base": wbase,  # kernel reads [0, nw); +1 harmless
+            "rec_len": slot_len,
+            "n_slots": nslots,
+        }
+        self._x_nrec = nrec
+
+    def _stage_xpose_host(self, buf):
+        """r1 host-side numpy builder (kept for A/B and layout tests)."""
         torch = self.t
         dev = self.eng.device
         gran = _env_int("DRAGNET_XGRAN", 32)

@@ -266,6 +266,29 @@ void scan_chunk(
               hipGetErrorString(err));
 }
 
+// Device-side wave-transpose: scatter length-sorted records into the
+// granule-interleaved layout (see xpose_build_kernel).
+void xpose_build(torch::Tensor data, torch::Tensor sstart,
+                 torch::Tensor slen, torch::Tensor wbase,
+                 int64_t n_slots, int64_t gran_log, torch::Tensor xb) {
+  CHECK_GPU(data);
+  CHECK_GPU(xb);
+  uint32_t nw = (uint32_t)(n_slots >> 6);
+  TORCH_CHECK(wbase.numel() >= (long)nw + 1, "wbase needs nw+1 rows");
+  uint32_t blocks = (nw + 3) / 4;  // 4 waves per 256-thread block
+  hipLaunchKernelGGL(xpose_build_kernel, dim3(blocks), dim3(256), 0,
+                     current_stream(),
+                     (const uint8_t*)data.data_ptr(),
+                     (const uint32_t*)sstart.data_ptr(),
+                     (const uint32_t*)slen.data_ptr(),
+                     (const unsigned long long*)wbase.data_ptr(),
+                     (uint32_t)n_slots, (int)gran_log,
+                     (uint8_t*)xb.data_ptr());
+  hipError_t err = hipGetLastError();
+  TORCH_CHECK(err == hipSuccess, "xpose_build launch failed: ",
+              hipGetErrorString(err));
+}
+
 // One-call zeroing of all scan state: a streaming step otherwise
 // issues ~10 separate torch .zero_() dispatches whose python+dispatch
 // overhead shows at 5 ms/step.
@@ -455,6 +478,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "dragnet_amd MI355X scan engine (gfx950 HIP kernels)";
   m.def("scan_chunk", &dn::scan_chunk, "fused NDJSON scan over one chunk");
   m.def("newline_index", &dn::newline_index, "device-side newline index");
+  m.def("xpose_build", &dn::xpose_build, "device-side wave transpose");
   m.def("agg_descs_host", &dn::agg_descs_host);
   m.def("extract_agg", &dn::extract_agg);
   m.def("extract_agg_async", &dn::extract_agg_async);

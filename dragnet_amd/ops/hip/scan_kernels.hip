@@ -1429,6 +1429,48 @@ __global__ void newline_write_kernel(const uint8_t* data,
   }
 }
 
+// Device-side wave-transpose builder: scatter each (length-sorted)
+// record's bytes into the granule-interleaved layout scan_kernel_x
+// consumes (byte p of slot r -> wbase[r/64] + (p/gran)*(64*gran) +
+// (r%64)*gran + p%gran).  One wave per 64-slot transposed wave: for a
+// fixed granule g the 64 lanes write 64 CONSECUTIVE granules (fully
+// coalesced 2 KiB stores); reads are per-record gathers (the cost the
+// transpose exists to pay once instead of every scan pass).
+__global__ void xpose_build_kernel(const uint8_t* __restrict__ data,
+                                   const uint32_t* __restrict__ sstart,
+                                   const uint32_t* __restrict__ slen,
+                                   const unsigned long long* __restrict__ wbase,
+                                   uint32_t n_slots, int gran_log,
+                                   uint8_t* __restrict__ xb) {
+  uint32_t w = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  uint32_t nw = n_slots >> 6;
+  if (w >= nw) return;
+  uint32_t lane = threadIdx.x & 63u;
+  uint32_t r = (w << 6) + lane;
+  uint32_t gran = 1u << gran_log;
+  unsigned long long base = wbase[w];
+  uint32_t gw = (uint32_t)((wbase[w + 1] - base) >> (6 + gran_log));
+  uint32_t len = slen[r];
+  if (len == 0xFFFFFFFFu) len = 0;  // padding slot
+  uint32_t start = sstart[r];
+  uint8_t* dst0 = xb + base + ((size_t)lane << gran_log);
+  for (uint32_t g = 0; g < gw; g++) {
+    uint8_t* dst = dst0 + ((size_t)g << (6 + gran_log));
+    uint32_t off = g << gran_log;
+    for (uint32_t k = 0; k < gran; k += 16) {
+      uint8_t tmp[16];
+#pragma unroll
+      for (int t = 0; t < 16; t++) {
+        uint32_t o = off + k + t;
+        tmp[t] = (o < len) ? data[start + o] : (uint8_t)'\n';
+      }
+      uint4 vv;
+      __builtin_memcpy(&vv, tmp, 16);
+      *reinterpret_cast<uint4*>(dst + k) = vv;
+    }
+  }
+}
+
 template <int XP>
 DEV void scan_kernel_body(char* smem, ScanArgs A);
 

@@ -511,3 +511,41 @@ def test_dense_overflow_falls_back(engines, tmp_path):
     c = cpu.scan([str(f)], [q])
     g = gpu.scan([str(f)], [q])
     assert_same(c, g)
+
+
+def test_xpose_device_vs_host_staging(engines, tmp_path):
+    """The device-side transposer (xpose_build_kernel)
+    must aggregate identically to the r1 host numpy builder."""
+    cpu, gpu = engines
+    from dragnet_amd.engine import plan as planmod
+    from dragnet_amd.engine.gpu import _ScanContext
+    from dragnet_amd.query import query_load
+    from dragnet_amd.tools.mktestdata import generate_lines
+    lines = list(generate_lines(25000, seed=123))
+    lines += [b"junk line\n", b'{"m": "' + b"y" * 700 + b'"}\n']
+    pool = b"".join(lines)
+    q = query_load(filter={"eq": ["req.method", "GET"]},
+                   breakdown_specs="req.method,res.statusCode")
+    path = tmp_path / "xpd.ndjson"
+    path.write_bytes(pool)
+    expected = cpu.scan([str(path)], [q]).aggregators[0].points()
+
+    results = {}
+    for mode in ("device", "host"):
+        if mode == "host":
+            os.environ["DRAGNET_XPOSE_HOST"] = "1"
+        try:
+            cplan = planmod.compile_plan([q])
+            ctx = _ScanContext(gpu, cplan, agg_slots=1 << 15,
+                               dict_slots=1 << 15,
+                               dict_data_cap=8 << 20)
+            ctx.stage_xpose(pool)
+            ctx.reset()
+            ctx.scan_xpose()
+            aggs, stages = ctx.finalize([q])
+            results[mode] = (aggs[0].points(),
+                             dict(stages)["json parser"])
+        finally:
+            os.environ.pop("DRAGNET_XPOSE_HOST", None)
+    assert results["device"][0] == expected
+    assert results["device"] == results["host"]

@@ -247,7 +247,11 @@ class FileDatasource(object):
 
         final = Aggregator(query)
         nerrors = []
-        use_gpu = os.environ.get("DRAGNET_INDEX_GPU") == "1"
+        # K7 policy: "1" forces the GPU columnar path, "0" forces
+        # SQLite; default "auto" uses the GPU only on a GPU engine and
+        # only above a measured row-count threshold (small tables are
+        # dominated by per-file setup; see profiles/r02_k7.md)
+        mode = os.environ.get("DRAGNET_INDEX_GPU", "auto")
         for path, _st in files:
             try:
                 iq = IndexQuerier(path)
@@ -255,10 +259,7 @@ class FileDatasource(object):
                 nerrors.append((path, str(e)))
                 continue
             try:
-                if use_gpu:
-                    partial = self._index_query_gpu(iq, eff)
-                else:
-                    partial = iq.run(eff)
+                partial = self._index_query(iq, eff, mode)
                 for p in partial.points():
                     final.write(p)
             except IndexError_ as e:
@@ -283,51 +284,49 @@ class FileDatasource(object):
         result.errors = nerrors
         return result
 
-    def _index_query_gpu(self, iq, query):
-        """K7: run an index query on the GPU by streaming the stored
-        rows as skinner points through the fused scan kernel (the
-        reference composes K7 from K2+K5 the same way).
+    def _index_query(self, iq, query, mode="auto"):
+        """K7: per-file index query, GPU-columnar when it pays.
 
-        The metric's rows are serialized to json-skinner NDJSON and fed
-        to the engine with the query's filter as the datasource filter;
-        the kernel re-filters, re-bucketizes and re-aggregates.
+        The GPU path reads the metric table's typed
+        columns through the native SQLite reader (_csink.read_columns)
+        and evaluates the query on-device over the raw columns
+        (columnar_query_kernel) — no NDJSON serialization or re-parse.
+        Reference semantics: lib/index-query.js:303-338 (SELECT ...
+        WHERE pred GROUP BY + re-aggregation).
         """
-        import json as _json
-
-        from ..index.sink import sqlite3_escape
+        if mode == "0" or self.engine().name != "gpu":
+            return iq.run(query)
         table = iq.find_metric(query)
+        if mode != "1":
+            n = iq.db.execute(
+                "SELECT COUNT(*) FROM %s" % table["table"]
+            ).fetchone()[0]
+            if n < _env_rows_threshold():
+                return iq.run(query)
 
+        from ..index import _csink
+        from ..index.sink import sqlite3_escape
         when = query.time_bounds_filter(table["datefield"]) \
             if table["datefield"] else None
         qfilter = None if table["ignore_filter"] else query.filter
         filt = krill.filter_and(qfilter, when)
 
-        cols = [p["name"] for p in table["params"]]
+        params = table["params"]
+        kinds = "".join(
+            "n" if ("aggr" in p or "date" in p) else "s"
+            for p in params)
         sql = "SELECT %s from %s" % (
-            ", ".join([sqlite3_escape(c) for c in cols] + ["value"]),
-            table["table"])
-
-        def rows_as_points():
-            buf = []
-            total = 0
-            for row in iq.db.execute(sql):
-                fields = dict(zip(cols, row[:-1]))
-                line = (_json.dumps(
-                    {"fields": fields, "value": row[-1]},
-                    separators=(",", ":")) + "\n").encode()
-                buf.append(line)
-                total += len(line)
-                if total >= (8 << 20):
-                    yield b"".join(buf)
-                    buf, total = [], 0
-            if buf:
-                yield b"".join(buf)
-
-        res = self.engine().scan(
-            files=[], queries=[query], ds_filter=filt,
-            data_format="json-skinner",
-            byte_source=rows_as_points())
-        return res.aggregators[0]
+            ", ".join([sqlite3_escape(p["name"]) for p in params]
+                      + ["value"]), table["table"])
+        cols, vals = _csink.read_columns(iq.filename, sql, kinds)
+        # stored strings are DECODED text; the kernel's string compare
+        # unescapes record spans, so literal backslashes would be
+        # misread — rare corner, served by SQLite instead
+        for k, c in zip(kinds, cols):
+            if k == "s" and b"\\" in c[0]:
+                return iq.run(query)
+        return self.engine().columnar_query(query, filt, params,
+                                            kinds, cols, vals)
 
     def index_read_points(self, metrics, interval="day"):
         """Emit every stored row of every index file as tagged points
@@ -361,6 +360,15 @@ class FileDatasource(object):
 def _esc(c):
     from ..index.sink import sqlite3_escape
     return sqlite3_escape(c)
+
+
+def _env_rows_threshold():
+    """Row count above which the GPU columnar index query pays
+    (measured: profiles/r02_k7.md)."""
+    try:
+        return int(os.environ.get("DRAGNET_INDEX_GPU_ROWS", 200000))
+    except ValueError:
+        return 200000
 
 
 def find_files(roots, counters=None):

@@ -43,10 +43,17 @@ int sqlite3_step(sqlite3_stmt *);
 int sqlite3_reset(sqlite3_stmt *);
 int sqlite3_finalize(sqlite3_stmt *);
 const char *sqlite3_errmsg(sqlite3 *);
+double sqlite3_column_double(sqlite3_stmt *, int);
+const unsigned char *sqlite3_column_text(sqlite3_stmt *, int);
+int sqlite3_column_bytes(sqlite3_stmt *, int);
+int sqlite3_column_type(sqlite3_stmt *, int);
 }
 
 #define SQLITE_OK 0
+#define SQLITE_ROW 100
 #define SQLITE_DONE 101
+#define SQLITE_NULL 5
+#define SQLITE_OPEN_READONLY 0x1
 #define SQLITE_OPEN_READWRITE 0x2
 #define SQLITE_OPEN_CREATE 0x4
 // SQLITE_TRANSIENT: sqlite copies the text before returning
@@ -243,6 +250,83 @@ struct CSink {
   ~CSink() { close(); }
 };
 
+// Columnar table extraction for the GPU index-query path (K7): run
+// `sql` (the metric table SELECT, value column LAST) and return typed
+// columns — f64 arrays for numeric columns, (blob, offs, lens) for
+// TEXT columns — at SQLite-C speed instead of per-row Python tuples.
+// kinds: one char per non-value column, 'n' = numeric, 's' = string.
+py::tuple read_columns(const std::string &path, const std::string &sql,
+                       const std::string &kinds) {
+  sqlite3 *db = nullptr;
+  if (sqlite3_open_v2(path.c_str(), &db, SQLITE_OPEN_READONLY,
+                      nullptr) != SQLITE_OK) {
+    std::string msg = db ? sqlite3_errmsg(db) : "open failed";
+    if (db) sqlite3_close(db);
+    throw std::runtime_error("sqlite open " + path + ": " + msg);
+  }
+  sqlite3_stmt *st = nullptr;
+  if (sqlite3_prepare_v2(db, sql.c_str(), -1, &st, nullptr) !=
+      SQLITE_OK) {
+    std::string msg = sqlite3_errmsg(db);
+    sqlite3_close(db);
+    throw std::runtime_error("sqlite prepare: " + msg);
+  }
+  const size_t nc = kinds.size();
+  struct SCol {
+    std::string blob;
+    std::vector<uint32_t> offs, lens;
+  };
+  std::vector<std::vector<double>> nums(nc);
+  std::vector<SCol> strs(nc);
+  std::vector<double> values;
+  int rc;
+  while ((rc = sqlite3_step(st)) == SQLITE_ROW) {
+    for (size_t c = 0; c < nc; c++) {
+      if (kinds[c] == 'n') {
+        nums[c].push_back(sqlite3_column_double(st, (int)c));
+      } else {
+        // column_text coerces stored integers the way TEXT affinity
+        // would; NULL -> empty string (rows are fully populated by
+        // the sink)
+        const unsigned char *t = sqlite3_column_text(st, (int)c);
+        int len = t ? sqlite3_column_bytes(st, (int)c) : 0;
+        SCol &sc = strs[c];
+        sc.offs.push_back((uint32_t)sc.blob.size());
+        sc.lens.push_back((uint32_t)len);
+        if (len) sc.blob.append((const char *)t, (size_t)len);
+      }
+    }
+    values.push_back(sqlite3_column_double(st, (int)nc));
+  }
+  sqlite3_finalize(st);
+  sqlite3_close(db);
+  if (rc != SQLITE_DONE)
+    throw std::runtime_error("sqlite step failed reading " + path);
+
+  py::list cols;
+  for (size_t c = 0; c < nc; c++) {
+    if (kinds[c] == 'n') {
+      py::array_t<double> a((py::ssize_t)nums[c].size());
+      std::memcpy(a.mutable_data(), nums[c].data(),
+                  nums[c].size() * sizeof(double));
+      cols.append(a);
+    } else {
+      SCol &sc = strs[c];
+      py::array_t<uint32_t> offs((py::ssize_t)sc.offs.size());
+      py::array_t<uint32_t> lens((py::ssize_t)sc.lens.size());
+      std::memcpy(offs.mutable_data(), sc.offs.data(),
+                  sc.offs.size() * sizeof(uint32_t));
+      std::memcpy(lens.mutable_data(), sc.lens.data(),
+                  sc.lens.size() * sizeof(uint32_t));
+      cols.append(py::make_tuple(py::bytes(sc.blob), offs, lens));
+    }
+  }
+  py::array_t<double> vals((py::ssize_t)values.size());
+  std::memcpy(vals.mutable_data(), values.data(),
+              values.size() * sizeof(double));
+  return py::make_tuple(cols, vals);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(_csink, m) {
@@ -255,4 +339,6 @@ PYBIND11_MODULE(_csink, m) {
       .def("insert_columnar", &CSink::insert_columnar)
       .def("commit", &CSink::commit)
       .def("close", &CSink::close);
+  m.def("read_columns", &read_columns,
+        "typed columnar extraction of one metric table");
 }

@@ -266,6 +266,92 @@ void scan_chunk(
               hipGetErrorString(err));
 }
 
+// Columnar index-query (K7): build the per-slot column descriptor
+// array (CPU bytes; caller copies to GPU) and launch the kernel.
+torch::Tensor col_descs_host(std::vector<int64_t> kinds,
+                             std::vector<torch::Tensor> nums,
+                             std::vector<torch::Tensor> soffs,
+                             std::vector<torch::Tensor> slens) {
+  int nf = (int)kinds.size();
+  auto out = torch::zeros({(long)(nf * sizeof(ColDesc))},
+                          torch::dtype(torch::kUInt8));
+  ColDesc* d = (ColDesc*)out.data_ptr();
+  for (int f = 0; f < nf; f++) {
+    d[f].kind = (int32_t)kinds[f];
+    d[f].num = kinds[f] == 1 ? (const double*)nums[f].data_ptr()
+                             : nullptr;
+    d[f].soff = kinds[f] == 2 ? (const uint32_t*)soffs[f].data_ptr()
+                              : nullptr;
+    d[f].slen = kinds[f] == 2 ? (const uint32_t*)slens[f].data_ptr()
+                              : nullptr;
+  }
+  return out;
+}
+
+void columnar_query(
+    torch::Tensor col_descs, torch::Tensor blob, torch::Tensor values,
+    int64_t nrows,
+    torch::Tensor field_sigs, torch::Tensor comp_slot,
+    int64_t nf_match, torch::Tensor prog_nodes,
+    torch::Tensor prog_bounds, torch::Tensor const_meta,
+    torch::Tensor const_dvals, torch::Tensor const_bytes,
+    torch::Tensor metric_rows, torch::Tensor synth_req,
+    torch::Tensor bd_rows, torch::Tensor bd_steps,
+    torch::Tensor table_descs,
+    torch::Tensor sd_state, torch::Tensor sd_hash, torch::Tensor sd_id,
+    torch::Tensor sd_off, torch::Tensor sd_len, torch::Tensor sd_data,
+    torch::Tensor sd_used, torch::Tensor sd_next,
+    torch::Tensor nd_state, torch::Tensor nd_bits, torch::Tensor nd_id,
+    torch::Tensor nd_next, torch::Tensor counters) {
+  CHECK_GPU(col_descs);
+  CHECK_GPU(blob);
+  ColArgs A;
+  A.nrows = (uint32_t)nrows;
+  A.values = (const double*)values.data_ptr();
+  A.blob = (const uint8_t*)blob.data_ptr();
+  A.cols = (const ColDesc*)col_descs.data_ptr();
+  A.P.field_sigs = (const uint64_t*)field_sigs.data_ptr();
+  A.P.nf = (int)field_sigs.numel();
+  A.P.comp_slot = (const int32_t*)comp_slot.data_ptr();
+  A.P.nf_match = (int)nf_match;
+  A.P.sig_bloom = 0;
+  A.P.prog_nodes = (const int32_t*)prog_nodes.data_ptr();
+  A.P.prog_bounds = (const int32_t*)prog_bounds.data_ptr();
+  A.P.const_meta = (const int32_t*)const_meta.data_ptr();
+  A.P.const_dvals = (const double*)const_dvals.data_ptr();
+  A.P.const_bytes = (const uint8_t*)const_bytes.data_ptr();
+  A.P.synth_slots = (const int32_t*)synth_req.data_ptr();  // unused
+  A.P.ns = 0;
+  A.P.metric_rows = (const int32_t*)metric_rows.data_ptr();
+  A.P.synth_req = (const int32_t*)synth_req.data_ptr();
+  A.P.bd_rows = (const int32_t*)bd_rows.data_ptr();
+  A.P.bd_steps = (const double*)bd_steps.data_ptr();
+  A.P.nm = (int)(metric_rows.numel() / 8);
+  A.P.value_slot = -1;
+  A.P.fields_slot = -1;
+  A.tables = (AggTable*)table_descs.data_ptr();
+  A.sdict = make_sdict(sd_state, sd_hash, sd_id, sd_off, sd_len,
+                       sd_data, sd_used, sd_next);
+  A.ndict = make_ndict(nd_state, nd_bits, nd_id, nd_next);
+  A.counters = (unsigned long long*)counters.data_ptr();
+
+  int nf = A.P.nf;
+  size_t lds = (size_t)nf * BLOCK * (8 + 4 + 4 + 1);
+  lds = (lds + 15) & ~(size_t)15;
+  lds += (size_t)LDS_CACHE * sizeof(LdsCacheEntry);
+  lds += (C_GLOBAL_N + (size_t)A.P.nm * CM_N) * 8;
+  lds += 64;
+  TORCH_CHECK(lds <= 160 * 1024, "columnar plan needs too much LDS");
+  uint32_t blocks = (uint32_t)((nrows + BLOCK - 1) / BLOCK);
+  if (blocks > 2048) blocks = 2048;
+  if (blocks == 0) return;
+  hipLaunchKernelGGL(columnar_query_kernel, dim3(blocks), dim3(BLOCK),
+                     lds, current_stream(), A);
+  hipError_t err = hipGetLastError();
+  TORCH_CHECK(err == hipSuccess, "columnar_query launch failed: ",
+              hipGetErrorString(err));
+}
+
 // Device-side wave-transpose: scatter length-sorted records into the
 // granule-interleaved layout (see xpose_build_kernel).
 void xpose_build(torch::Tensor data, torch::Tensor sstart,
@@ -479,6 +565,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scan_chunk", &dn::scan_chunk, "fused NDJSON scan over one chunk");
   m.def("newline_index", &dn::newline_index, "device-side newline index");
   m.def("xpose_build", &dn::xpose_build, "device-side wave transpose");
+  m.def("col_descs_host", &dn::col_descs_host);
+  m.def("columnar_query", &dn::columnar_query,
+        "K7: query over uploaded index columns");
   m.def("agg_descs_host", &dn::agg_descs_host);
   m.def("extract_agg", &dn::extract_agg);
   m.def("extract_agg_async", &dn::extract_agg_async);

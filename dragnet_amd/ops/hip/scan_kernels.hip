@@ -1429,6 +1429,210 @@ __global__ void newline_write_kernel(const uint8_t* data,
   }
 }
 
+// -------------------------------------------------------------------
+// Columnar index-query kernel (K7): evaluate a query directly over an
+// index table's typed columns (reference lib/index-query.js:303-338
+// runs SELECT ... WHERE pred GROUP BY; here the predicate + bucketize
+// + hash-aggregate run over uploaded columns — no NDJSON round trip).
+// Rows are weighted points: weight = the stored SUM(value).
+
+struct ColDesc {
+  int32_t kind;  // 0 = missing for every row, 1 = numeric, 2 = string
+  int32_t pad_;
+  const double* num;     // [nrows] (kind 1)
+  const uint32_t* soff;  // [nrows] into the shared blob (kind 2)
+  const uint32_t* slen;
+};
+
+struct ColArgs {
+  uint32_t nrows;
+  const double* values;  // [nrows] row weights
+  const uint8_t* blob;   // concatenated string payloads
+  const ColDesc* cols;   // [P.nf] per plan slot
+  PlanView P;
+  AggTable* tables;
+  StrDict sdict;
+  NumDict ndict;
+  unsigned long long* counters;
+};
+
+__launch_bounds__(BLOCK, 4)
+__global__ void columnar_query_kernel(ColArgs A) {
+  extern __shared__ __attribute__((aligned(16))) char smemc[];
+  const PlanView& P = A.P;
+  const int nf = P.nf;
+  size_t off = 0;
+  double* fv_num = reinterpret_cast<double*>(smemc + off);
+  off += (size_t)nf * BLOCK * sizeof(double);
+  uint32_t* fv_soff = reinterpret_cast<uint32_t*>(smemc + off);
+  off += (size_t)nf * BLOCK * sizeof(uint32_t);
+  uint32_t* fv_slen = reinterpret_cast<uint32_t*>(smemc + off);
+  off += (size_t)nf * BLOCK * sizeof(uint32_t);
+  uint8_t* fv_type = reinterpret_cast<uint8_t*>(smemc + off);
+  off += (size_t)nf * BLOCK * sizeof(uint8_t);
+  off = (off + 15) & ~(size_t)15;
+  LdsCacheEntry* cache = reinterpret_cast<LdsCacheEntry*>(smemc + off);
+  off += (size_t)LDS_CACHE * sizeof(LdsCacheEntry);
+  unsigned long long* lcnt =
+      reinterpret_cast<unsigned long long*>(smemc + off);
+  const int NCNT = C_GLOBAL_N + P.nm * CM_N;
+
+  for (int i = threadIdx.x; i < LDS_CACHE; i += BLOCK) {
+    cache[i].hash = 0;
+    cache[i].metric = 0;
+    for (int k = 0; k < MAX_KEY; k++) cache[i].key[k] = 0;
+    cache[i].count = 0.0;
+  }
+  for (int i = threadIdx.x; i < NCNT; i += BLOCK) lcnt[i] = 0;
+  __syncthreads();
+
+  FV fv;
+  fv.type = fv_type; fv.soff = fv_soff; fv.slen = fv_slen;
+  fv.num = fv_num; fv.tid = threadIdx.x;
+  Bytes BV;
+  BV.mem = A.blob;
+  BV.bias = 0;
+
+  const uint32_t stride = gridDim.x * BLOCK;
+  for (uint32_t r = blockIdx.x * BLOCK + threadIdx.x; r < A.nrows;
+       r += stride) {
+    atomicAdd(&lcnt[C_LINES], 1ull);
+    atomicAdd(&lcnt[C_PARSED], 1ull);
+    for (int f = 0; f < nf; f++) {
+      const ColDesc& c = A.cols[f];
+      if (c.kind == 1)
+        fv.set(f, T_NUM, 0, 0, c.num[r]);
+      else if (c.kind == 2)
+        fv.set(f, T_STR, c.soff[r], c.slen[r], 0.0);
+      else
+        fv.set(f, T_MISSING, 0, 0, 0.0);
+    }
+    // program 0: the combined (query ∧ time-bounds) filter
+    {
+      int keep = eval_predicate(P, BV, fv, 0);
+      if (keep != 1) {
+        if (keep == -1) atomicAdd(&lcnt[C_DS_FAILEDEVAL], 1ull);
+        else atomicAdd(&lcnt[C_DS_FILTERED], 1ull);
+        continue;
+      }
+    }
+    double weight = A.values[r];
+    for (int m = 0; m < P.nm; m++) {
+      const int32_t* M = &P.metric_rows[m * 8];
+      unsigned long long* mc = &lcnt[C_GLOBAL_N + m * CM_N];
+      atomicAdd(&mc[CM_AGG_IN], 1ull);
+
+      uint32_t key[MAX_KEY];
+      int nk = M[1];
+      bool drop = false, overflow = false;
+      for (int bi = 0; bi < nk; bi++) {
+        const int32_t* B = &P.bd_rows[(M[2] + bi) * 4];
+        double step = P.bd_steps[M[2] + bi];
+        int slot = B[1];  // columnar plans use kind-0 refs only
+        uint8_t t = fv.get_type(slot);
+        double num = fv.get_num(slot);
+        uint32_t so = fv.get_soff(slot), sl = fv.get_slen(slot);
+        uint32_t code;
+        if (B[2] != BUCKET_NONE) {
+          if (t == T_STR) {
+            num = js_to_number(BV, so, sl);
+            if (!(num == num) || num == __builtin_inf() ||
+                num == -__builtin_inf()) {
+              drop = true;
+              break;
+            }
+            t = T_NUM;
+          }
+          if (t != T_NUM) { drop = true; break; }
+          long long ord;
+          if (B[2] == BUCKET_P2) {
+            if (!(num >= 1.0)) ord = 0;
+            else {
+              uint64_t bits = __double_as_longlong(num);
+              ord = (long long)((bits >> 52) & 0x7FF) - 1023 + 1;
+            }
+          } else {
+            ord = (long long)__builtin_floor(num / step);
+          }
+          if (ord >= -(long long)ORD_BIAS &&
+              ord < (long long)ORD_BIAS) {
+            code = make_code(TAG_ORD, (uint32_t)(ord + ORD_BIAS));
+          } else {
+            uint32_t id = intern_number(A.ndict, (double)ord);
+            if (id == 0xFFFFFFFFu) { overflow = true; break; }
+            code = make_code(TAG_NUM, id);
+          }
+        } else if (t == T_MISSING) {
+          code = make_code(TAG_SPECIAL, SPECIAL_UNDEF);
+        } else if (t == T_NUM) {
+          uint32_t id = intern_number(A.ndict, num);
+          if (id == 0xFFFFFFFFu) { overflow = true; break; }
+          code = make_code(TAG_NUM, id);
+        } else {  // T_STR
+          uint32_t id = intern_string(A.sdict, BV, so, sl);
+          if (id == 0xFFFFFFFFu) { overflow = true; break; }
+          code = make_code(TAG_STR, id);
+        }
+#pragma unroll
+        for (int kk = 0; kk < MAX_KEY; kk++)
+          if (kk == bi) key[kk] = code;
+      }
+      if (overflow) { atomicAdd(&lcnt[C_OVERFLOW], 1ull); continue; }
+      if (drop) { atomicAdd(&mc[CM_NONNUMERIC], 1ull); continue; }
+#pragma unroll
+      for (int k = 0; k < MAX_KEY; k++)
+        if (k >= nk) key[k] = 0;
+
+      uint64_t kh = mix64((uint64_t)m * 0x9E3779B97F4A7C15ull + 1);
+      for (int k = 0; k < MAX_KEY; k++) kh = mix64(kh ^ key[k]);
+      if (kh == 0) kh = 1;
+      bool cached = false;
+      uint32_t ci = (uint32_t)kh & (LDS_CACHE - 1);
+      for (int attempt = 0; attempt < 8; attempt++) {
+        unsigned long long prev = atomicCAS(
+            (unsigned long long*)&cache[ci].hash, 0ull,
+            (unsigned long long)kh);
+        if (prev == 0) {
+          cache[ci].metric = m;
+          for (int k = 0; k < MAX_KEY; k++) cache[ci].key[k] = key[k];
+          atomicAdd(&cache[ci].count, weight);
+          cached = true;
+          break;
+        }
+        if (prev == (unsigned long long)kh && cache[ci].metric == m) {
+          bool same = true;
+          for (int k = 0; k < MAX_KEY; k++)
+            if (cache[ci].key[k] != key[k]) { same = false; break; }
+          if (same) {
+            atomicAdd(&cache[ci].count, weight);
+            cached = true;
+            break;
+          }
+        }
+        ci = (ci + 1) & (LDS_CACHE - 1);
+      }
+      if (!cached) {
+        if (!agg_add(A.tables[m], key, nk, weight))
+          atomicAdd(&lcnt[C_OVERFLOW], 1ull);
+      }
+    }
+  }
+
+  __syncthreads();
+  for (int i = threadIdx.x; i < LDS_CACHE; i += BLOCK) {
+    if (cache[i].hash != 0) {
+      if (!agg_add(A.tables[cache[i].metric], cache[i].key,
+                   P.metric_rows[cache[i].metric * 8 + 1],
+                   cache[i].count))
+        atomicAdd(&lcnt[C_OVERFLOW], 1ull);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < NCNT; i += BLOCK) {
+    if (lcnt[i]) atomicAdd(&A.counters[i], lcnt[i]);
+  }
+}
+
 // Device-side wave-transpose builder: scatter each (length-sorted)
 // record's bytes into the granule-interleaved layout scan_kernel_x
 // consumes (byte p of slot r -> wbase[r/64] + (p/gran)*(64*gran) +

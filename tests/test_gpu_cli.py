@@ -116,3 +116,34 @@ def test_cli_sharded_backend_gpu(dn, fixture_tree, monkeypatch):
     b = dn("scan", "-b", "req.method", "fl")
     assert a.code == 0 and b.code == 0, (a.err, b.err)
     assert a.out == b.out
+
+
+def test_cli_query_index_gpu_columnar_bounds(dn, fixture_tree,
+                                             tmp_path, monkeypatch):
+    """K7 columnar with time-bounded queries over an interval tree
+    (__dn_ts date column in the WHERE) and with weighted rows: GPU
+    columnar == SQLite."""
+    idx = str(tmp_path / "idxb")
+    monkeypatch.setenv("DRAGNET_ENGINE", "gpu")
+    r = dn("datasource-add", "tree", "--path=" + fixture_tree,
+           "--index-path=" + idx, "--time-field=time",
+           "--time-format=%Y/%m-%d")
+    assert r.code == 0, r.err
+    r = dn("metric-add", "tree", "m", "-b",
+           "host,req.method,latency[aggr=quantize]")
+    assert r.code == 0, r.err
+    r = dn("build", "tree")
+    assert r.code == 0, r.err
+    cases = [("-b", "host", "--after", "2014-05-01",
+              "--before", "2014-05-03"),
+             ("-b", "req.method,latency[aggr=quantize]",
+              "--after", "2014-05-02", "--before", "2014-05-05"),
+             ("-b", "host,req.method")]
+    for case in cases:
+        monkeypatch.setenv("DRAGNET_INDEX_GPU", "0")
+        sql_res = dn("query", *case, "tree")
+        monkeypatch.setenv("DRAGNET_INDEX_GPU", "1")
+        gpu_res = dn("query", *case, "tree")
+        assert sql_res.code == 0 and gpu_res.code == 0, \
+            (case, gpu_res.err)
+        assert gpu_res.out == sql_res.out, case

@@ -363,12 +363,20 @@ def _esc(c):
 
 
 def _env_rows_threshold():
-    """Row count above which the GPU columnar index query pays
-    (measured: profiles/r02_k7.md)."""
+    """Row count above which the GPU columnar index query would pay.
+
+    Measured (profiles/r02_k7.md): extracting columns out of the
+    SQLite B-tree alone costs ~2.5x what SQLite's own WHERE+GROUP BY
+    takes end-to-end, so the GPU path loses at EVERY size while
+    indexes are SQLite files — auto mode therefore never selects it
+    (threshold = infinity) and DRAGNET_INDEX_GPU=1 forces it."""
+    v = os.environ.get("DRAGNET_INDEX_GPU_ROWS")
+    if v is None:
+        return float("inf")
     try:
-        return int(os.environ.get("DRAGNET_INDEX_GPU_ROWS", 200000))
+        return int(v)
     except ValueError:
-        return 200000
+        return float("inf")
 
 
 def find_files(roots, counters=None):

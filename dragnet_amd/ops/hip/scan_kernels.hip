@@ -1301,10 +1301,16 @@ struct LdsCacheEntry {
 
 constexpr uint32_t NL_SEG = 2048;
 
-// SWAR newline mask: high bit set in each byte lane equal to '\n'
+// SWAR newline mask: high bit set in each byte lane equal to '\n'.
+// EXACT per-byte form: the classic (x-0x01..)&~x&0x80.. zero test is
+// only exact as a boolean — a borrow out of a true-zero byte false-
+// positives the next byte when it is 0x01 (i.e. the byte 0x0B right
+// after a real '\n'; caught by the envelope-crossing fuzzer on binary
+// garbage lines).  This form has no cross-byte carries.
 DEV uint32_t nl_mask32(uint32_t w) {
-  uint32_t x = w ^ 0x0A0A0A0Au;
-  return (x - 0x01010101u) & ~x & 0x80808080u;
+  uint32_t x = w ^ 0x0A0A0A0Au;            // zero byte where '\n'
+  uint32_t y = (x & 0x7F7F7F7Fu) + 0x7F7F7F7Fu;  // hi set iff low7!=0
+  return ~(y | x | 0x7F7F7F7Fu);           // hi set iff byte==0
 }
 
 // byte-validity mask for a word at byte address wpos over [start, n)

@@ -129,7 +129,11 @@ def test_cli_query_index_gpu_columnar_bounds(dn, fixture_tree,
            "--index-path=" + idx, "--time-field=time",
            "--time-format=%Y/%m-%d")
     assert r.code == 0, r.err
+    # time-bounded index queries need an explicit date breakdown in
+    # the metric (findMetric requires a date param; reference
+    # lib/index-query.js:190-203)
     r = dn("metric-add", "tree", "m", "-b",
+           "ts[date,field=time,aggr=lquantize,step=3600],"
            "host,req.method,latency[aggr=quantize]")
     assert r.code == 0, r.err
     r = dn("build", "tree")
@@ -138,6 +142,8 @@ def test_cli_query_index_gpu_columnar_bounds(dn, fixture_tree,
               "--before", "2014-05-03"),
              ("-b", "req.method,latency[aggr=quantize]",
               "--after", "2014-05-02", "--before", "2014-05-05"),
+             ("-b", "ts[date,field=time,aggr=lquantize,step=3600],host",
+              "--after", "2014-05-02", "--before", "2014-05-03"),
              ("-b", "host,req.method")]
     for case in cases:
         monkeypatch.setenv("DRAGNET_INDEX_GPU", "0")

@@ -59,6 +59,14 @@ class GpuEngine(object):
         self.device = device if device is not None else \
             torch.device("cuda", torch.cuda.current_device())
         self.chunk_bytes = _env_int("DRAGNET_CHUNK_MB", 256) * 1024 * 1024
+        # chunk addressing is uint32 on-device: offsets/positions must
+        # fit 4 GiB per chunk (arbitrarily large INPUTS stream through
+        # many chunks; only a single RECORD larger than the chunk is
+        # rejected, at scan time)
+        if self.chunk_bytes > (3 << 30):
+            raise ValueError(
+                "DRAGNET_CHUNK_MB too large: device chunk addressing "
+                "is 32-bit (max 3072 MB per chunk)")
 
     # ---- public engine interface ----
 

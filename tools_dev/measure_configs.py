@@ -133,7 +133,7 @@ def main():
             "--time-field=time", "--index-path=" + idx_root], env=env)
 
     # ---- config #2: dn scan -b req.method, 10 GB ----
-    for readers in (8, 16):
+    for readers in (8, 16, 24, 32):
         env["DRAGNET_READERS"] = str(readers)
         wall, out, err = run_dn(
             ["scan", "-b", "req.method", "d10"], env=env, timing=True)
@@ -166,7 +166,7 @@ def main():
     nrec100 = nrec_base * copies100
     run_dn(["datasource-add", "d100", "--path=" + data100,
             "--time-field=time"], env=env)
-    env["DRAGNET_READERS"] = "16"
+    env["DRAGNET_READERS"] = os.environ.get("MC_READERS", "24")
     wall, out, err = run_dn(
         ["scan", "-f", '{"eq": ["req.method", "GET"]}',
          "-b", "req.method,res.statusCode", "d100"],

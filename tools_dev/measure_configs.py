@@ -97,6 +97,26 @@ def run_dn(args, env=None, timing=False):
     return wall, r.stdout, r.stderr
 
 
+def api_scan_time(data_dir, filter_=None, breakdowns="req.method",
+                  readers=24):
+    """In-process engine-level scan timing (excludes interpreter/torch
+    startup, which the CLI wall numbers include)."""
+    os.environ["DRAGNET_READERS"] = str(readers)
+    from dragnet_amd.config import Datasource
+    from dragnet_amd.datasource.file import FileDatasource
+    from dragnet_amd.engine import get_engine
+    from dragnet_amd.query import query_load
+    ds = Datasource(name="t", backend="file", path=data_dir,
+                    time_field="time")
+    fd = FileDatasource(ds, engine=get_engine())
+    q = query_load(filter=filter_, breakdown_specs=breakdowns)
+    t0 = time.time()
+    r = fd.scan(q)
+    dt = time.time() - t0
+    n = dict(r.stages)["json parser"]["ninputs"]
+    return dt, n
+
+
 def main():
     out_path = os.path.join(REPO, "gpurun_out", "configs_r02.json")
     results = []
@@ -154,6 +174,17 @@ def main():
         log(json.dumps(res))
         results.append(res)
 
+    # engine-level timing for #2 (no startup), reader sweep
+    for readers in (8, 16, 24, 32):
+        dt, nrec = api_scan_time(data10, readers=readers)
+        res = {"config": "#2 engine-level scan", "readers": readers,
+               "bytes": bytes10, "records": nrec,
+               "scan_s": round(dt, 3),
+               "gb_per_sec": round(bytes10 / dt / 1e9, 2),
+               "recs_per_sec": round(nrec / dt, 0)}
+        log(json.dumps(res))
+        results.append(res)
+
     # ---- config #3: filter + 2-field breakdown, as large as fits ---
     target3 = min(budget - bytes10,
                   int(float(os.environ.get("MC_T100_GB", 100)) * GB))
@@ -182,6 +213,17 @@ def main():
         "gb_per_sec": round(bytes100 / (wall - req) / 1e9, 2),
         "recs_per_sec": round(nrec100 / (wall - req), 0),
     }
+    log(json.dumps(res))
+    results.append(res)
+    dt, nrec = api_scan_time(
+        data100, filter_={"eq": ["req.method", "GET"]},
+        breakdowns="req.method,res.statusCode",
+        readers=int(os.environ.get("MC_READERS", "24")))
+    res = {"config": "#3 engine-level scan",
+           "bytes": bytes100, "records": nrec,
+           "scan_s": round(dt, 3),
+           "gb_per_sec": round(bytes100 / dt / 1e9, 2),
+           "recs_per_sec": round(nrec / dt, 0)}
     log(json.dumps(res))
     results.append(res)
 

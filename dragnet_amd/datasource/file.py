@@ -325,6 +325,11 @@ class FileDatasource(object):
         for k, c in zip(kinds, cols):
             if k == "s" and b"\\" in c[0]:
                 return iq.run(query)
+        # two breakdowns reading the same source column can't share a
+        # kernel-plan name — rare; served by SQLite
+        srcs = [b.get("field", b["name"]) for b in query.breakdowns]
+        if len(set(srcs)) != len(srcs):
+            return iq.run(query)
         return self.engine().columnar_query(query, filt, params,
                                             kinds, cols, vals)
 

@@ -137,12 +137,18 @@ class GpuEngine(object):
         dev = self.device
         from ..query import QueryConfig
 
-        # kernel plan: each breakdown reads its STORED column by name
-        # as a plain field (dates are already materialized seconds);
-        # the caller's query does the date/bucket decode
+        # kernel plan: each breakdown reads the column keyed by its
+        # FIELD name — the reference's deserializeRow reads
+        # row[escape(field.field)] (lib/index-query.js:395-400), so a
+        # renamed date breakdown (field != name) reads a column that
+        # was never selected -> undefined -> dropped by the
+        # bucketizer.  Stored columns are keyed by NAME, so mapping
+        # the kernel slot by field reproduces exactly that; the
+        # caller's query does the date/bucket decode.
         kq_bds = []
         for b in query.breakdowns:
-            nb = {"name": b["name"], "field": b["name"]}
+            src = b.get("field", b["name"])
+            nb = {"name": src, "field": src}
             if "aggr" in b:
                 nb["aggr"] = b["aggr"]
                 if "step" in b:
@@ -389,9 +395,9 @@ class _ScanContext(object):
         each chunk is cut at its last newline and the tail is carried
         into the head of the other buffer while the GPU works on the
         previous chunk.  Regular-file chunks are filled with parallel
-        preadv calls (page-cache reads are single-core bound at
-        ~15-25 GB/s; the 8-reader default measured 34.5 GB/s
-        end-to-end on a warm 4 GB file, 3x the 4-reader rate)."""
+        preadv calls (page-cache/tmpfs reads are single-core bound
+        at ~10 GB/s; the r2 microbench sweep measured 52/67/55 GB/s
+        at 8/16/32 readers — 16 is the default)."""
         import concurrent.futures as cf
         import stat as _stat
 
@@ -405,7 +411,7 @@ class _ScanContext(object):
         views = [memoryview(p.numpy()) for p in pins]
         evs = [None, None]
         pool = cf.ThreadPoolExecutor(
-            max_workers=_env_int("DRAGNET_READERS", 8))
+            max_workers=_env_int("DRAGNET_READERS", 16))
 
         def pread_full(fd, mv, off):
             """preadv until mv is full (a single preadv may legally
@@ -431,7 +437,7 @@ class _ScanContext(object):
                 return seq_file.readinto(view[at:at + want]) or 0
             if want < (8 << 20):
                 return pread_full(fd, view[at:at + want], fpos)
-            nsec = _env_int("DRAGNET_READERS", 8)
+            nsec = _env_int("DRAGNET_READERS", 16)
             sec = (want + nsec - 1) // nsec
             futs = []
             for s in range(0, want, sec):
@@ -532,17 +538,41 @@ class _ScanContext(object):
             pool.shutdown(wait=False)
 
     def _run_pinned(self, pin, n):
-        """H2D + kernels for pin[:n] (already newline-terminated)."""
+        """H2D + kernels for pin[:n] (already newline-terminated).
+
+        Ping-pong DEVICE buffers with the H2D on a dedicated copy
+        stream: chunk k+1's copy overlaps chunk k's kernels, so the
+        steady state is max(fill, copy, kernels) instead of their sum
+        (the serial form measured 12-15 GB/s end-to-end vs the 67 GB/s
+        the parallel readers deliver; see profiles/r02 notes).
+        Returns the H2D-complete event — the caller must wait on it
+        before refilling this pinned buffer."""
         torch = self.t
         padded = _pad(n)
         pin[n:padded] = 10
-        dev_data = self._dev_data
-        dev_data[:padded].copy_(pin[:padded], non_blocking=True)
+        if not hasattr(self, "_fs_dev"):
+            self._fs_copy_stream = torch.cuda.Stream(
+                device=self.eng.device)
+            self._fs_dev = [self._dev_data,
+                            torch.empty_like(self._dev_data)]
+            self._fs_done = [torch.cuda.Event(), torch.cuda.Event()]
+            self._fs_idx = 0
+        idx = self._fs_idx
+        self._fs_idx ^= 1
+        dev_data = self._fs_dev[idx]
+        main = torch.cuda.current_stream(self.eng.device)
+        cs = self._fs_copy_stream
+        # last kernel reader of this device buffer must finish first
+        cs.wait_event(self._fs_done[idx])
         ev = torch.cuda.Event()
-        ev.record()
+        with torch.cuda.stream(cs):
+            dev_data[:padded].copy_(pin[:padded], non_blocking=True)
+            ev.record(cs)
+        main.wait_event(ev)
         self.eng.ops.newline_index(dev_data, 0, n, self._segs,
                                    self._pos, self._nlines)
         self._scan_call(dev_data, 0)
+        self._fs_done[idx].record(main)
         return ev
 
     # ---- resident-pool path (bench / repeated scans) ----

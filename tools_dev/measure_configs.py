@@ -227,6 +227,28 @@ def main():
     log(json.dumps(res))
     results.append(res)
 
+    # ---- config #5 shape (single-node leg): sharded-backend scan
+    # over the same corpus; world=1 here — the 8-rank DP + RCCL merge
+    # leg is the driver's SCALE run (bench.py is exactly that shape)
+    run_dn(["datasource-add", "d100s", "--backend=sharded",
+            "--path=" + data100, "--time-field=time"], env=env)
+    wall, out, err = run_dn(
+        ["scan", "-f", '{"eq": ["req.method", "GET"]}',
+         "-b", "req.method,res.statusCode", "d100s"],
+        env=env, timing=True)
+    req = 0.0
+    for ln in err.splitlines():
+        if "require:" in ln:
+            req = float(ln.split()[-1].rstrip("s"))
+    res = {
+        "config": "#5 sharded-backend scan (world=1 leg)",
+        "bytes": bytes100, "records": nrec100,
+        "wall_s": round(wall, 3), "require_s": round(req, 3),
+        "gb_per_sec": round(bytes100 / (wall - req) / 1e9, 2),
+    }
+    log(json.dumps(res))
+    results.append(res)
+
     # ---- config #4: build + query with the muskie index ----
     wall_b, out, err = run_dn(
         ["build", "--index-config",

@@ -723,7 +723,7 @@ class _ScanContext(object):
             return self._stage_xpose_host(buf)
         torch = self.t
         dev = self.eng.device
-        gran = _env_int("DRAGNET_XGRAN", 32)
+        gran = _env_int("DRAGNET_XGRAN", 64)
         glog = gran.bit_length() - 1
         if not hasattr(self, "_resident"):
             if buf is None:
@@ -773,7 +773,7 @@ class _ScanContext(object):
         """r1 host-side numpy builder (kept for A/B and layout tests)."""
         torch = self.t
         dev = self.eng.device
-        gran = _env_int("DRAGNET_XGRAN", 32)
+        gran = _env_int("DRAGNET_XGRAN", 64)
         xb, wave_base, slot_len, nslots, n = _build_xpose_layout(
             buf, gran)
         self._x = {

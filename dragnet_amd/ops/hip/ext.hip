@@ -219,10 +219,10 @@ void scan_chunk(
   if (A.xn_slots > 0) {
     blocks = (A.xn_slots + BLOCK - 1) / BLOCK;
     if (blocks > 2048) blocks = 2048;
-    // 32B granules measured fastest (less padding, tighter rows):
-    // 1101 vs 1079 (64B) vs 1023 (128B) M rec/s on one box
+    // r2 sweep on the device-built layout: 64B granules fastest
+    // (1457-1462 vs 1436 (32B) vs 1400 (128B) M rec/s)
     const char* xg_env = getenv("DRAGNET_XGRAN");
-    int xg = xg_env ? atoi(xg_env) : 32;
+    int xg = xg_env ? atoi(xg_env) : 64;
     if (xg == 32)
       hipLaunchKernelGGL((scan_kernel_x<4, 5>), dim3(blocks),
                          dim3(BLOCK), lds, current_stream(), A);

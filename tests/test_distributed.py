@@ -263,3 +263,55 @@ def test_sharded_build_partitioned(fixture_tree, tmp_path):
     via_local = fd1.query(q, interval="day").aggregators[0].points()
     assert via_dist == via_local
     assert via_dist  # non-empty
+
+
+@pytest.mark.timeout(180)
+def test_launch_job_builder(fixture_tree, tmp_path):
+    """dn-launch (the Manta job-builder analog): --dry-run prints the
+    job definition; --gpus 2 actually runs a 2-rank distributed scan
+    through torch.distributed.run whose rank-0 output equals a
+    single-process scan."""
+    import json as _json
+    import subprocess
+    import sys as _sys
+
+    r = subprocess.run(
+        [_sys.executable, "-m", "dragnet_amd.launch", "--gpus", "2",
+         "--dry-run", "scan", "-b", "req.method", "src"],
+        capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    job = _json.loads(r.stdout)
+    assert job["nprocPerNode"] == 2
+    assert job["phases"][0]["exec"].endswith(
+        "-m dragnet_amd.cli scan -b req.method src")
+
+    # real 2-rank run (gloo on CPU), output == single-process scan
+    cfgfile = str(tmp_path / "rc.json")
+    env = dict(os.environ, DRAGNET_CONFIG=cfgfile,
+               DRAGNET_ENGINE="cpu")
+    from dragnet_amd import config as mod_config
+    cfg = mod_config.DragnetConfig()
+    cfg.datasource_add(mod_config.Datasource(
+        name="src", backend="sharded", path=fixture_tree))
+    mod_config.save_config(cfg, cfgfile)
+    r2 = subprocess.run(
+        [_sys.executable, "-m", "dragnet_amd.launch", "--gpus", "2",
+         "scan", "-b", "req.method", "src"],
+        capture_output=True, text=True, env=env, timeout=150)
+    assert r2.returncode == 0, r2.stderr[-2000:]
+
+    cfg2 = mod_config.DragnetConfig()
+    cfg2.datasource_add(mod_config.Datasource(
+        name="src", backend="file", path=fixture_tree))
+    mod_config.save_config(cfg2, str(tmp_path / "rc2.json"))
+    env1 = dict(env, DRAGNET_CONFIG=str(tmp_path / "rc2.json"))
+    r1_ = subprocess.run(
+        [_sys.executable, "-m", "dragnet_amd.cli",
+         "scan", "-b", "req.method", "src"],
+        capture_output=True, text=True, env=env1)
+    assert r1_.returncode == 0, r1_.stderr
+    # gloo prints connection banners on stdout; drop them
+    dist_out = "".join(
+        ln for ln in r2.stdout.splitlines(keepends=True)
+        if not ln.startswith("[Gloo]"))
+    assert dist_out == r1_.stdout

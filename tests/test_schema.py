@@ -81,3 +81,18 @@ def test_json_streamer(tmp_path):
     assert r.returncode == 0
     assert r.stdout.strip() == b"3"
     assert b"warn:" in r.stderr
+
+
+def test_memwatch(tmp_path):
+    import subprocess
+    import sys as _sys
+    p = subprocess.Popen([_sys.executable, "-c",
+                          "import time; x='a'*(40<<20); time.sleep(1.2)"])
+    r = subprocess.run(
+        [_sys.executable, "-m", "dragnet_amd.tools.memwatch",
+         str(p.pid), "0.2"], capture_output=True, text=True)
+    p.wait()
+    assert r.returncode == 0
+    rss, vsz = map(int, r.stdout.split())
+    assert rss > 40 * 1024  # saw the 40 MB allocation (KB units)
+    assert vsz >= rss

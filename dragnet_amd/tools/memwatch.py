@@ -12,11 +12,14 @@ import time
 
 
 def sample(pid):
-    """(rss_kb, vsz_kb) from /proc, or None if the process is gone."""
+    """(rss_kb, vsz_kb) from /proc, or None if the process is gone
+    (a zombie counts as gone: its memory is released)."""
     try:
         with open("/proc/%d/status" % pid) as f:
             rss = vsz = 0
             for line in f:
+                if line.startswith("State:") and "Z" in line.split()[1]:
+                    return None
                 if line.startswith("VmRSS:"):
                     rss = int(line.split()[1])
                 elif line.startswith("VmSize:"):

@@ -315,3 +315,66 @@ def test_launch_job_builder(fixture_tree, tmp_path):
         ln for ln in r2.stdout.splitlines(keepends=True)
         if not ln.startswith("[Gloo]"))
     assert dist_out == r1_.stdout
+
+
+def _worker_w4(rank, world, port, files, out_q):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["DRAGNET_ENGINE"] = "cpu"
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    from dragnet_amd.distributed import (init_process_group,
+                                         merge_tables_tensor)
+    from dragnet_amd.engine.cpu import CpuEngine
+    from dragnet_amd.query import query_load
+    init_process_group(backend="gloo")
+    # rank 3 gets NO files (empty-shard edge); others round-robin 3-way
+    mine = [] if rank == 3 else \
+        [f for i, f in enumerate(files) if i % 3 == rank]
+    q1 = query_load(breakdown_specs="operation,res.statusCode")
+    q2 = query_load(filter={"eq": ["req.method", "NOSUCH"]},
+                    breakdown_specs="host")  # matches nothing anywhere
+    res = CpuEngine().scan(mine, [q1, q2])
+    m1 = merge_tables_tensor(res.aggregators[0], q1)
+    m2 = merge_tables_tensor(res.aggregators[1], q2)
+    out_q.put((rank, m1.points(), m2.points()))
+    import torch.distributed as dist
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_world4_empty_shard_and_empty_tables(fixture_tree):
+    """Dense merge at world 4 with one EMPTY rank shard and a second
+    query whose filter matches nothing on any rank (all-empty
+    all_gather): every rank must converge on the single-process
+    answer."""
+    files = []
+    for root, _dirs, names in os.walk(fixture_tree):
+        for n in sorted(names):
+            files.append(os.path.join(root, n))
+    files.sort()
+    from dragnet_amd.engine.cpu import CpuEngine
+    from dragnet_amd.query import query_load
+    q1 = query_load(breakdown_specs="operation,res.statusCode")
+    single = CpuEngine().scan(files, [q1])
+    expected = single.aggregators[0].points()
+
+    ctx = mp.get_context("spawn")
+    out_q = ctx.Queue()
+    world = 4
+    procs = [ctx.Process(target=_worker_w4,
+                         args=(r, world, 29539, files, out_q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [out_q.get(timeout=120) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    for _rank, p1, p2 in results:
+        assert p1 == expected
+        assert p2 == []  # empty table merges to empty everywhere

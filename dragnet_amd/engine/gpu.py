@@ -405,6 +405,17 @@ class _ScanContext(object):
 
         torch = self.t
         cap = self.eng.chunk_bytes
+        # adaptive chunk size: big corpora amortize per-chunk overhead
+        # with 1 GiB chunks (+~10% measured at 24 readers); small scans
+        # keep the cheap-startup default
+        if (self._pinned is None
+                and os.environ.get("DRAGNET_CHUNK_MB") is None):
+            try:
+                total = sum(os.stat(p).st_size for p in files)
+            except OSError:
+                total = 0
+            if total > (16 << 30):
+                cap = self.eng.chunk_bytes = 1 << 30
         if self._pinned is None:
             self._ensure_buffers(_pad(cap))
         pins = [self._pinned,

@@ -63,7 +63,11 @@ def init_process_group(backend=None):
     if world == 1:
         return None
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # DRAGNET_DIST_BACKEND=gloo lets N ranks rehearse on one GPU
+        # (RCCL refuses duplicate devices); production defaults to
+        # RCCL when CUDA/HIP devices are visible
+        backend = os.environ.get("DRAGNET_DIST_BACKEND") or \
+            ("nccl" if torch.cuda.is_available() else "gloo")
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29517")
     if backend == "nccl":

@@ -140,3 +140,49 @@ def test_fuzz_differential(seed, tmp_path):
         cs = dict(c.stages)["json parser"]
         gs = dict(g.stages)["json parser"]
         assert gs == cs, desc
+
+
+def test_fuzz_json_skinner_weighted(engines, tmp_path):
+    """Weighted json-skinner re-aggregation fuzz: random POINT streams
+    (weights, literal dotted keys, junk lines) through both engines —
+    the distributed-reduce path (reference reduce phase consumes
+    exactly this format)."""
+    cpu, gpu = engines
+    for seed in (11, 12, 13, 14):
+        rng = random.Random(seed)
+        lines = []
+        for i in range(3000):
+            r = rng.random()
+            if r < 0.75:
+                fields = {}
+                for _ in range(rng.randrange(1, 5)):
+                    fields[rng.choice(KEYS)] = rng.choice(
+                        ["GET", "PUT", 200, 404, "h%d" % (i % 9),
+                         None, 2 ** (i % 12), "26"])
+                w = rng.choice([1, 2, 3, 5, 0.5, 1.25])
+                lines.append(json.dumps(
+                    {"fields": fields, "value": w}).encode())
+            elif r < 0.85:
+                # invalid points: missing fields/value, bad types
+                lines.append(rng.choice([
+                    b'{"fields": {"a": 1}}',
+                    b'{"value": 3}',
+                    b'{"fields": {"a": 1}, "value": "x"}',
+                    b'{"fields": {"a": 1}, "value": true}',
+                    b'not json',
+                    b'42',
+                ]))
+            else:
+                lines.append(json.dumps(
+                    {"fields": {"a.b": rng.randrange(5)},
+                     "value": rng.randrange(1, 4)}).encode())
+        path = tmp_path / ("sk%d.ndjson" % seed)
+        path.write_bytes(b"\n".join(lines) + b"\n")
+        for qi in range(3):
+            q = rand_query(rng)
+            c = cpu.scan([str(path)], [q], data_format="json-skinner")
+            g = gpu.scan([str(path)], [q], data_format="json-skinner")
+            assert g.aggregators[0].points() == \
+                c.aggregators[0].points(), (seed, qi, q.filter)
+            assert dict(g.stages)["json parser"] == \
+                dict(c.stages)["json parser"], (seed, qi)

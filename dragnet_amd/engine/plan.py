@@ -71,30 +71,34 @@ def _mix64(x):
     return x
 
 
-def comp_hash(b):
-    """Hash of one path component over zero-padded 8-byte words —
-    computable on-device with SWAR window loads (scan_kernels.hip
-    scan_key_sig must produce identical values)."""
-    h = 0x9E3779B97F4A7C15
-    for k in range(0, max(len(b), 1), 8):
+SIG_LENK = 0xFF51AFD7ED558CCD
+
+
+def comp_into(sig, b):
+    """Fold one path component's bytes into the chained signature:
+    zero-padded 8-byte little-endian words, one mix per word, plus a
+    length-finalization mix — computable on-device with inline SWAR
+    window folds during the key scan (scan_kernels.hip scan_key_sig /
+    sig_comp_finish must produce identical values)."""
+    for k in range(0, len(b), 8):
         w = int.from_bytes(b[k:k + 8].ljust(8, b"\0"), "little")
-        h = _mix64(h ^ w)
-    return _mix64(h ^ len(b))
+        sig = _mix64(sig ^ w)
+    return _mix64(sig ^ ((len(b) * SIG_LENK) & MASK64))
 
 
 def lit_sig(name):
     """Signature of a whole literal key (dots as plain bytes) at the
     top level — the companion-slot match target for json mode."""
-    return _mix64(FNV_OFFSET ^ comp_hash(name.encode("utf-8")))
+    return comp_into(FNV_OFFSET, name.encode("utf-8"))
 
 
 def path_sig(path):
-    """Signature of a dotted path: a chained component hash.  Dots
+    """Signature of a dotted path: chained component folds.  Dots
     split components, so a literal "a.b" key and nested a->b produce
     the same signature (the aggregation lookup rule)."""
     sig = FNV_OFFSET
     for comp in path.split("."):
-        sig = _mix64(sig ^ comp_hash(comp.encode("utf-8")))
+        sig = comp_into(sig, comp.encode("utf-8"))
     return sig
 
 

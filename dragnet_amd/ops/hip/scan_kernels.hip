@@ -256,23 +256,35 @@ DEV bool scan_string_fast(CursorT<BS>& c, uint32_t& off_out,
   }
 }
 
-// Component hash over zero-padded 8-byte words (mirrors
-// plan.comp_hash; any overread is masked away and stays inside the
-// padded buffer).
+// Streamed signature chaining (mirrors plan.py comp_into): key bytes
+// fold DIRECTLY into the path signature as zero-padded 8-byte words —
+// one mix per word plus one length-finalization mix, with the words
+// hashed inline during the key scan (no re-load pass, one mix fewer
+// per component than the hash-then-chain scheme).
+constexpr uint64_t SIG_LENK = 0xFF51AFD7ED558CCDull;
+
+DEV inline uint64_t sig_word(uint64_t sig, uint64_t w) {
+  return mix64(sig ^ w);
+}
+DEV inline uint64_t sig_fin(uint64_t sig, uint32_t len) {
+  return mix64(sig ^ ((uint64_t)len * SIG_LENK));
+}
+
+// Fold the unhashed remainder [hashed, term) of the current component
+// (8-byte strides stay aligned to comp_s), then finalize with the
+// component length.
 template <class BS>
-DEV uint64_t comp_hash_span(BS B, uint32_t s, uint32_t e) {
-  uint64_t h = 0x9E3779B97F4A7C15ull;
-  uint32_t len = e - s;
-  uint32_t k = 0;
-  do {
-    uint64_t w = B.load8(s + k);
-    uint32_t rem = len - k;
-    if (rem < 8)
-      w = (rem == 0) ? 0ull : (w & ((~0ull) >> (8 * (8 - rem))));
-    h = mix64(h ^ w);
+DEV uint64_t sig_comp_finish(BS B, uint64_t sig, uint32_t hashed,
+                             uint32_t comp_s, uint32_t term) {
+  uint32_t k = hashed;
+  while (k < term) {
+    uint64_t w = B.load8(k);
+    uint32_t rem = term - k;
+    if (rem < 8) w &= (~0ull) >> (8 * (8 - rem));
+    sig = mix64(sig ^ w);
     k += 8;
-  } while (k < len);
-  return mix64(h ^ (uint64_t)len);
+  }
+  return sig_fin(sig, term - comp_s);
 }
 
 // Scan a KEY (cursor after the opening quote): SWAR windows to the
@@ -292,28 +304,39 @@ template <class BS>
 DEV int scan_key_sig(CursorT<BS>& c, uint64_t parent, uint64_t& sig_out,
                      bool dot_splits) {
   uint32_t p = c.pos, end = c.end;
-  uint32_t comp_s = p;
+  uint32_t comp_s = p;   // current component start
+  uint32_t hashed = p;   // bytes below this are already folded in
   uint64_t sig = parent;
   while (true) {
     while (p + 8 <= end) {
       uint64_t w = c.B.load8(p);
-      uint64_t m = str_special_mask(w) |
-                   hz8(w ^ 0x2E2E2E2E2E2E2E2Eull);
-      if (m == 0) { p += 8; continue; }
+      // json mode hashes dots as plain component bytes, so they must
+      // NOT break the window (alignment from comp_s is load-bearing)
+      uint64_t m = str_special_mask(w);
+      if (dot_splits) m |= hz8(w ^ 0x2E2E2E2E2E2E2E2Eull);
+      if (m == 0) {
+        if (p == hashed) {  // full window: fold inline
+          sig = sig_word(sig, w);
+          hashed = p + 8;
+        }
+        p += 8;
+        continue;
+      }
       p += ((uint32_t)__ffsll((unsigned long long)m) - 1) >> 3;
       break;
     }
     if (p >= end) return 0;
     uint8_t b = c.B.at(p);
     if (b == '"') {
-      sig_out = mix64(sig ^ comp_hash_span(c.B, comp_s, p));
+      sig_out = sig_comp_finish(c.B, sig, hashed, comp_s, p);
       c.pos = p + 1;
       return 1;
     }
     if (b == '.') {
       if (dot_splits) {
-        sig = mix64(sig ^ comp_hash_span(c.B, comp_s, p));
+        sig = sig_comp_finish(c.B, sig, hashed, comp_s, p);
         comp_s = p + 1;
+        hashed = p + 1;
       }
       p++;
       continue;

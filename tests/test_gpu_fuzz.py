@@ -142,12 +142,17 @@ def test_fuzz_differential(seed, tmp_path):
         assert gs == cs, desc
 
 
-def test_fuzz_json_skinner_weighted(engines, tmp_path):
+def test_fuzz_json_skinner_weighted(tmp_path):
     """Weighted json-skinner re-aggregation fuzz: random POINT streams
     (weights, literal dotted keys, junk lines) through both engines —
     the distributed-reduce path (reference reduce phase consumes
     exactly this format)."""
-    cpu, gpu = engines
+    import torch
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from dragnet_amd.engine.cpu import CpuEngine
+    from dragnet_amd.engine.gpu import GpuEngine
+    cpu, gpu = CpuEngine(), GpuEngine()
     for seed in (11, 12, 13, 14):
         rng = random.Random(seed)
         lines = []

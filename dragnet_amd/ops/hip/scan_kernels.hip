@@ -188,6 +188,24 @@ struct CursorT {
   DEV bool eof() const { return pos >= end; }
   DEV uint8_t peek() { return byte_at(pos); }
   DEV uint8_t next() { return byte_at(pos++); }
+  // 8 bytes starting at p, served from the 16B register window
+  // (refills like byte_at; may read window slack past `end` — host
+  // padding guarantees it)
+  DEV uint64_t word_at(uint32_t p) {
+#ifdef DN_DIRECT_BYTES
+    return B.load8(p);
+#else
+    uint32_t d = p - wbase;
+    if (d > 8u) {
+      wbase = p;
+      win = B.load8(p);
+      win2 = B.load8(p + 8);
+      d = 0;
+    }
+    if (d == 0) return win;
+    return (win >> (8u * d)) | (win2 << (64u - 8u * d));
+#endif
+  }
   DEV void skip_ws() {
     while (pos < end) {
       uint8_t b = byte_at(pos);
@@ -389,6 +407,50 @@ DEV double scale10(double v, long ex) {
 
 struct NumOut { double v; bool ok; };
 
+// SWAR digit-run scan: consume the run of ASCII digits at c.pos,
+// folding into mant (10^ndig positional, capped at 19 significant
+// digits with the overflow counted in extra).  Returns the number of
+// digits consumed.  The classifier flags the FIRST non-digit exactly
+// (carries out of a non-digit byte only corrupt LATER bytes, which
+// are beyond the stop point by construction).
+template <class BS>
+DEV int scan_digit_run(CursorT<BS>& c, uint64_t& mant, int& ndig,
+                       int& extra) {
+  int total = 0;
+  while (!c.eof()) {
+    uint64_t w = c.word_at(c.pos);
+    uint64_t t = w ^ 0x3030303030303030ull;
+    uint64_t nd = ((t + 0x7676767676767676ull) | t)
+                  & 0x8080808080808080ull;
+    uint32_t n = nd ? (((uint32_t)__ffsll((long long)nd) - 1) >> 3)
+                    : 8u;
+    uint32_t avail = c.end - c.pos;
+    if (n > avail) n = avail;
+    if (n == 0) break;
+    if (n == 8 && ndig + 8 <= 19) {
+      // 3-multiply 8-digit fold (first digit in the LOW byte = most
+      // significant)
+      uint64_t pairs = (t * ((10ull << 8) + 1)) >> 8;
+      uint64_t quads = ((pairs & 0x00FF00FF00FF00FFull)
+                        * ((100ull << 16) + 1)) >> 16;
+      uint64_t v8 = ((quads & 0x0000FFFF0000FFFFull)
+                     * ((10000ull << 32) + 1)) >> 32;
+      mant = mant * 100000000ull + (uint32_t)v8;
+      ndig += 8;
+    } else {
+      for (uint32_t k = 0; k < n; k++) {
+        uint32_t d = (uint32_t)(w >> (8 * k)) & 0xFF;
+        if (ndig < 19) { mant = mant * 10u + (d - '0'); ndig++; }
+        else extra++;
+      }
+    }
+    c.pos += n;
+    total += (int)n;
+    if (n < 8) break;
+  }
+  return total;
+}
+
 template <class BS>
 DEV NumOut parse_json_number(CursorT<BS>& c) {
   NumOut out; out.ok = false; out.v = 0.0;
@@ -403,27 +465,18 @@ DEV NumOut parse_json_number(CursorT<BS>& c) {
     c.pos++; ndig = 1;
     if (!c.eof()) { uint8_t nb = c.peek(); if (nb >= '0' && nb <= '9') return out; }
   } else if (b >= '1' && b <= '9') {
-    while (!c.eof()) {
-      uint8_t d = c.peek();
-      if (d < '0' || d > '9') break;
-      c.pos++;
-      if (ndig < 19) { mant = mant * 10u + (d - '0'); ndig++; }
-      else extra_exp++;
-    }
+    scan_digit_run(c, mant, ndig, extra_exp);
   } else {
     return out;
   }
   // fraction
   if (!c.eof() && c.peek() == '.') {
     c.pos++;
-    int fdig = 0;
-    while (!c.eof()) {
-      uint8_t d = c.peek();
-      if (d < '0' || d > '9') break;
-      c.pos++;
-      if (ndig < 19) { mant = mant * 10u + (d - '0'); ndig++; extra_exp--; }
-      fdig++;
-    }
+    int fdig = 0, fex = 0;
+    int nd0 = ndig;
+    fdig = scan_digit_run(c, mant, ndig, fex);
+    extra_exp -= (ndig - nd0);  // significant fraction digits
+    (void)fex;  // truncated fraction digits shift nothing
     if (fdig == 0) return out;
   }
   // exponent

@@ -203,6 +203,7 @@ struct CursorT {
       d = 0;
     }
     if (d == 0) return win;
+    if (d == 8u) return win2;  // a 64-bit shift by 64 would be UB
     return (win >> (8u * d)) | (win2 << (64u - 8u * d));
 #endif
   }

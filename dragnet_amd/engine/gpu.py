@@ -696,7 +696,7 @@ class _ScanContext(object):
             nlines if nlines is not None else self._nlines,
             first_start,
             self.field_sigs, self.comp_slot, self.cplan.nf_match,
-            self.cplan.sig_bloom,
+            self.cplan.sig_bloom, self.cplan.fields_parent_sig,
             self.prog_nodes, self.prog_bounds,
             self.const_meta, self.const_dvals, self.const_bytes,
             self.synth_slots, self.cplan.n_synth,

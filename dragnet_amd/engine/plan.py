@@ -72,6 +72,7 @@ def _mix64(x):
 
 
 SIG_LENK = 0xFF51AFD7ED558CCD
+SIG_LIT_MARK = 0xC2B2AE3D27D4EB4F  # fields-root literal-dotted marker
 
 
 def comp_into(sig, b):
@@ -360,8 +361,8 @@ def compile_plan(queries, ds_filter=None, time_field=None,
     nf_match = len(fields.paths)
     comp_slot = np.full(max(nf_match, 1), -1, dtype=np.int32)
     comp_sigs = []
-    if data_format != "json-skinner" and \
-            os.environ.get("DRAGNET_NO_COMPANIONS") != "1":
+    no_companions = os.environ.get("DRAGNET_NO_COMPANIONS") == "1"
+    if data_format != "json-skinner" and not no_companions:
         for i, (pth, raw) in enumerate(fields.paths):
             if not raw and "." in pth:
                 phys = nf_match + len(comp_sigs)
@@ -371,6 +372,26 @@ def compile_plan(queries, ds_filter=None, time_field=None,
                         "(incl. dotted-key companions)" % MAX_FIELDS)
                 comp_slot[i] = phys
                 comp_sigs.append(lit_sig(pth))
+    elif data_format == "json-skinner" and not no_companions:
+        # skinner: a literal dotted KEY at point.fields top level
+        # chains (dot-splits) to the SAME sig as the plucked nested
+        # path — correct for the aggregation readout (points.lookup is
+        # literal-first) but it must stay INVISIBLE to predicates and
+        # synthetic sources (krill pluck).  The device marks
+        # fields-root keys that contained a dot by folding
+        # SIG_LIT_MARK into their sig, landing them in the companion
+        # slot; breakdown readout consults the companion first,
+        # everything else reads the primary (pluck) slot.
+        for i, (pth, raw) in enumerate(fields.paths):
+            if not raw and "." in pth:
+                phys = nf_match + len(comp_sigs)
+                if phys >= MAX_FIELDS:
+                    raise PlanError(
+                        "query references more than %d fields "
+                        "(incl. dotted-key companions)" % MAX_FIELDS)
+                comp_slot[i] = phys
+                comp_sigs.append(_mix64(
+                    path_sig(prefix + pth) ^ SIG_LIT_MARK))
 
     programs = np.array(prog_nodes, dtype=np.int32).reshape(-1, 4)
     bounds = np.array(prog_bounds, dtype=np.int32).reshape(-1, 2)
@@ -402,6 +423,10 @@ def compile_plan(queries, ds_filter=None, time_field=None,
              np.array(comp_sigs, dtype=np.uint64)])
     plan.value_slot = value_slot
     plan.fields_slot = fields_slot
+    # parent sig of point.fields children (the device detects
+    # fields-root literal-dotted keys by comparing parents)
+    fps = path_sig("fields") if data_format == "json-skinner" else 0
+    plan.fields_parent_sig = fps - (1 << 64) if fps >= (1 << 63) else fps
     plan.ds_prog = ds_prog
     plan.data_format = data_format
     return plan

@@ -143,6 +143,8 @@ struct PlanView {
   int nm;
   int value_slot;               // json-skinner weight slot (-1 if json)
   int fields_slot;              // json-skinner "fields" presence (-1)
+  uint64_t fields_parent_sig;   // path_sig("fields") in skinner mode
+                                // (fields-root literal-dotted detect)
 };
 
 struct ScanArgs {

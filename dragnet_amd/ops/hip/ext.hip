@@ -119,7 +119,8 @@ void scan_chunk(
     torch::Tensor data, torch::Tensor nl_pos, torch::Tensor nlines_dev,
     int64_t first_start,
     torch::Tensor field_sigs, torch::Tensor comp_slot,
-    int64_t nf_match, int64_t sig_bloom, torch::Tensor prog_nodes,
+    int64_t nf_match, int64_t sig_bloom, int64_t fields_parent_sig,
+    torch::Tensor prog_nodes,
     torch::Tensor prog_bounds, torch::Tensor const_meta,
     torch::Tensor const_dvals, torch::Tensor const_bytes,
     torch::Tensor synth_slots, int64_t n_synth,
@@ -153,6 +154,7 @@ void scan_chunk(
   A.P.comp_slot = (const int32_t*)comp_slot.data_ptr();
   A.P.nf_match = (int)nf_match;
   A.P.sig_bloom = (uint64_t)sig_bloom;
+  A.P.fields_parent_sig = (uint64_t)fields_parent_sig;
   A.P.prog_nodes = (const int32_t*)prog_nodes.data_ptr();
   A.P.prog_bounds = (const int32_t*)prog_bounds.data_ptr();
   A.P.const_meta = (const int32_t*)const_meta.data_ptr();
@@ -315,6 +317,7 @@ void columnar_query(
   A.P.comp_slot = (const int32_t*)comp_slot.data_ptr();
   A.P.nf_match = (int)nf_match;
   A.P.sig_bloom = 0;
+  A.P.fields_parent_sig = 0;
   A.P.prog_nodes = (const int32_t*)prog_nodes.data_ptr();
   A.P.prog_bounds = (const int32_t*)prog_bounds.data_ptr();
   A.P.const_meta = (const int32_t*)const_meta.data_ptr();

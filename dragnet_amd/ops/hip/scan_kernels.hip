@@ -281,6 +281,8 @@ DEV bool scan_string_fast(CursorT<BS>& c, uint32_t& off_out,
 // hashed inline during the key scan (no re-load pass, one mix fewer
 // per component than the hash-then-chain scheme).
 constexpr uint64_t SIG_LENK = 0xFF51AFD7ED558CCDull;
+// fields-root literal-dotted marker (plan.py SIG_LIT_MARK)
+constexpr uint64_t SIG_LIT_MARK = 0xC2B2AE3D27D4EB4Full;
 
 DEV inline uint64_t sig_word(uint64_t sig, uint64_t w) {
   return mix64(sig ^ w);
@@ -326,6 +328,7 @@ DEV int scan_key_sig(CursorT<BS>& c, uint64_t parent, uint64_t& sig_out,
   uint32_t comp_s = p;   // current component start
   uint32_t hashed = p;   // bytes below this are already folded in
   uint64_t sig = parent;
+  bool saw_dot = false;
   while (true) {
     while (p + 8 <= end) {
       uint64_t w = c.B.load8(p);
@@ -349,13 +352,14 @@ DEV int scan_key_sig(CursorT<BS>& c, uint64_t parent, uint64_t& sig_out,
     if (b == '"') {
       sig_out = sig_comp_finish(c.B, sig, hashed, comp_s, p);
       c.pos = p + 1;
-      return 1;
+      return saw_dot ? 2 : 1;  // 2 = key contained a split dot
     }
     if (b == '.') {
       if (dot_splits) {
         sig = sig_comp_finish(c.B, sig, hashed, comp_s, p);
         comp_s = p + 1;
         hashed = p + 1;
+        saw_dot = true;
       }
       p++;
       continue;
@@ -717,8 +721,16 @@ DEV bool parse_record(BS BV, uint32_t start, uint32_t end,
   // parent signature via chained component hashes (plan.path_sig)
   auto parse_key = [&](uint64_t parent, bool root, uint64_t& sig_out) -> bool {
     if (c.eof() || c.next() != '"') return false;
-    return scan_key_sig(c, root ? FNV_OFFSET : parent, sig_out,
-                        dot_splits) != 0;
+    int r = scan_key_sig(c, root ? FNV_OFFSET : parent, sig_out,
+                         dot_splits);
+    if (r == 0) return false;
+    // skinner: a fields-ROOT key containing a dot is a LITERAL dotted
+    // key — addressable by the aggregation lookup (literal-first) but
+    // invisible to krill pluck; fold the marker so it captures into
+    // the companion slot (plan.py SIG_LIT_MARK)
+    if (r == 2 && !root && parent == P.fields_parent_sig)
+      sig_out = mix64(sig_out ^ SIG_LIT_MARK);
+    return true;
   };
 
   // Main loop: parse values iteratively.

@@ -75,6 +75,15 @@ SIG_LENK = 0xFF51AFD7ED558CCD
 SIG_LIT_MARK = 0xC2B2AE3D27D4EB4F  # fields-root literal-dotted marker
 
 
+def _sig_mix(x):
+    """5-op bijective mix used ONLY for path signatures (mirrors
+    scan_kernels.hip sig_mix)."""
+    x &= MASK64
+    x ^= x >> 32
+    x = (x * 0xD6E8FEB86659FD93) & MASK64
+    return x ^ (x >> 32)
+
+
 def comp_into(sig, b):
     """Fold one path component's bytes into the chained signature:
     zero-padded 8-byte little-endian words, one mix per word, plus a
@@ -83,8 +92,8 @@ def comp_into(sig, b):
     sig_comp_finish must produce identical values)."""
     for k in range(0, len(b), 8):
         w = int.from_bytes(b[k:k + 8].ljust(8, b"\0"), "little")
-        sig = _mix64(sig ^ w)
-    return _mix64(sig ^ ((len(b) * SIG_LENK) & MASK64))
+        sig = _sig_mix(sig ^ w)
+    return _sig_mix(sig ^ ((len(b) * SIG_LENK) & MASK64))
 
 
 def lit_sig(name):
@@ -390,7 +399,7 @@ def compile_plan(queries, ds_filter=None, time_field=None,
                         "query references more than %d fields "
                         "(incl. dotted-key companions)" % MAX_FIELDS)
                 comp_slot[i] = phys
-                comp_sigs.append(_mix64(
+                comp_sigs.append(_sig_mix(
                     path_sig(prefix + pth) ^ SIG_LIT_MARK))
 
     programs = np.array(prog_nodes, dtype=np.int32).reshape(-1, 4)

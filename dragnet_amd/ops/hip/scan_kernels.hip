@@ -281,14 +281,24 @@ DEV bool scan_string_fast(CursorT<BS>& c, uint32_t& off_out,
 // hashed inline during the key scan (no re-load pass, one mix fewer
 // per component than the hash-then-chain scheme).
 constexpr uint64_t SIG_LENK = 0xFF51AFD7ED558CCDull;
+
+// 5-op bijective mix for the SIGNATURE chain only (dict/table hashing
+// keeps the full murmur finalizer): xor-fold, odd-multiply, xor-fold.
+// Bijective (hi recoverable, then lo), and the final fold feeds the
+// bloom's low 6 bits from the whole state.  plan.py _sig_mix mirrors.
+DEV inline uint64_t sig_mix(uint64_t x) {
+  x ^= x >> 32;
+  x *= 0xD6E8FEB86659FD93ull;
+  return x ^ (x >> 32);
+}
 // fields-root literal-dotted marker (plan.py SIG_LIT_MARK)
 constexpr uint64_t SIG_LIT_MARK = 0xC2B2AE3D27D4EB4Full;
 
 DEV inline uint64_t sig_word(uint64_t sig, uint64_t w) {
-  return mix64(sig ^ w);
+  return sig_mix(sig ^ w);
 }
 DEV inline uint64_t sig_fin(uint64_t sig, uint32_t len) {
-  return mix64(sig ^ ((uint64_t)len * SIG_LENK));
+  return sig_mix(sig ^ ((uint64_t)len * SIG_LENK));
 }
 
 // Fold the unhashed remainder [hashed, term) of the current component
@@ -302,7 +312,7 @@ DEV uint64_t sig_comp_finish(BS B, uint64_t sig, uint32_t hashed,
     uint64_t w = B.load8(k);
     uint32_t rem = term - k;
     if (rem < 8) w &= (~0ull) >> (8 * (8 - rem));
-    sig = mix64(sig ^ w);
+    sig = sig_mix(sig ^ w);
     k += 8;
   }
   return sig_fin(sig, term - comp_s);
@@ -729,7 +739,7 @@ DEV bool parse_record(BS BV, uint32_t start, uint32_t end,
     // invisible to krill pluck; fold the marker so it captures into
     // the companion slot (plan.py SIG_LIT_MARK)
     if (r == 2 && !root && parent == P.fields_parent_sig)
-      sig_out = mix64(sig_out ^ SIG_LIT_MARK);
+      sig_out = sig_mix(sig_out ^ SIG_LIT_MARK);
     return true;
   };
 

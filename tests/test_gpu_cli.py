@@ -227,8 +227,9 @@ def test_cli_stdin_scan_gpu(tmp_path):
     subprocess.run([_sys.executable, "-m", "dragnet_amd.cli",
                     "datasource-add", "stdin", "--path=/dev/stdin"],
                    env=env, check=True)
+    filt = "{ \"eq\": [ \"req.method\", \"GET\" ] }"
     argv = [_sys.executable, "-m", "dragnet_amd.cli", "scan",
-            "-f", '"'"'{ "eq": [ "req.method", "GET" ] }'"'"',
+            "-f", filt,
             "-b", "req.method,res.statusCode", "stdin"]
     out = {}
     for engine in ("cpu", "gpu"):

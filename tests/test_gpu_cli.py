@@ -231,11 +231,17 @@ def test_cli_stdin_scan_gpu(tmp_path):
     argv = [_sys.executable, "-m", "dragnet_amd.cli", "scan",
             "-f", filt,
             "-b", "req.method,res.statusCode", "stdin"]
+    # redirect stdin FROM A FILE (the reference's stdin fixtures do
+    # the same: under a pipe /dev/stdin is a FIFO, which fs-find
+    # skips — regular files and char devices only)
+    pf = tmp_path / "pool.ndjson"
+    pf.write_bytes(pool)
     out = {}
     for engine in ("cpu", "gpu"):
         e = dict(env, DRAGNET_ENGINE=engine)
-        r = subprocess.run(argv, input=pool, capture_output=True,
-                           env=e)
+        with open(pf, "rb") as fh:
+            r = subprocess.run(argv, stdin=fh, capture_output=True,
+                               env=e)
         assert r.returncode == 0, (engine, r.stderr[-1500:])
         out[engine] = r.stdout
     assert out["gpu"] == out["cpu"]

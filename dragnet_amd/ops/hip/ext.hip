@@ -191,7 +191,7 @@ void scan_chunk(
   lds = (lds + 15) & ~(size_t)15;
   lds += (size_t)LDS_CACHE * sizeof(LdsCacheEntry);
   lds += (C_GLOBAL_N + (size_t)A.P.nm * CM_N) * 8;
-  lds += (size_t)MAX_SYNTH * BLOCK * 8;  // synthetic values
+  lds += (size_t)A.P.ns * BLOCK * 8;  // synthetic values (actual count)
   lds += (size_t)6 /*SIG_DEPTH*/ * BLOCK * 8;  // parse sig stack
   lds += 64;  // slack
   TORCH_CHECK(lds <= 160 * 1024, "plan needs too much LDS: ", lds);

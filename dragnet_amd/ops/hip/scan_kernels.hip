@@ -2103,7 +2103,10 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
   const int NCNT = C_GLOBAL_N + P.nm * CM_N;
   off += (size_t)NCNT * sizeof(unsigned long long);
   double* synth_lds = reinterpret_cast<double*>(smem + off);
-  off += (size_t)MAX_SYNTH * BLOCK * sizeof(double);
+  // sized by the ACTUAL synthetic count: the flagship plan has zero
+  // synthetics and the unconditional MAX_SYNTH reservation (16 KB)
+  // was costing a whole extra block of occupancy per CU
+  off += (size_t)P.ns * BLOCK * sizeof(double);
   uint64_t* sig_lds = reinterpret_cast<uint64_t*>(smem + off);
   off += (size_t)SIG_DEPTH * BLOCK * sizeof(uint64_t);
   off = (off + 15) & ~(size_t)15;

@@ -187,7 +187,7 @@ void scan_chunk(
   }
 
   int nf = A.P.nf;
-  size_t lds = (size_t)nf * BLOCK * (8 + 4 + 4 + 1);
+  size_t lds = (size_t)nf * BLOCK * (4 + 4 + 1);
   lds = (lds + 15) & ~(size_t)15;
   lds += (size_t)LDS_CACHE * sizeof(LdsCacheEntry);
   lds += (C_GLOBAL_N + (size_t)A.P.nm * CM_N) * 8;
@@ -339,7 +339,7 @@ void columnar_query(
   A.counters = (unsigned long long*)counters.data_ptr();
 
   int nf = A.P.nf;
-  size_t lds = (size_t)nf * BLOCK * (8 + 4 + 4 + 1);
+  size_t lds = (size_t)nf * BLOCK * (4 + 4 + 1);
   lds = (lds + 15) & ~(size_t)15;
   lds += (size_t)LDS_CACHE * sizeof(LdsCacheEntry);
   lds += (C_GLOBAL_N + (size_t)A.P.nm * CM_N) * 8;

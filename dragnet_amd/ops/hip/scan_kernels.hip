@@ -390,22 +390,35 @@ DEV int scan_key_sig(CursorT<BS>& c, uint64_t parent, uint64_t& sig_out,
 // per-record extracted field values, stored in LDS (SoA, [field][tid])
 
 struct FV {
+  // (soff, slen) double as the VALUE bits for T_NUM slots — a slot is
+  // either a span (str/obj/arr) or a number, never both, and every
+  // get_num consumer is type-gated.  The union halves the per-field
+  // LDS footprint (9 B/field/thread), which buys occupancy: LDS is
+  // the block-count limiter on this kernel.
   uint8_t* type;   // nf * BLOCK
-  uint32_t* soff;  // nf * BLOCK
-  uint32_t* slen;  // nf * BLOCK
-  double* num;     // nf * BLOCK
+  uint32_t* soff;  // nf * BLOCK; T_NUM: low 32 bits of the double
+  uint32_t* slen;  // nf * BLOCK; T_NUM: high 32 bits
   int tid;
   DEV void set(int f, uint8_t t, uint32_t off, uint32_t len, double n) {
     type[f * BLOCK + tid] = t;
-    soff[f * BLOCK + tid] = off;
-    slen[f * BLOCK + tid] = len;
-    num[f * BLOCK + tid] = n;
+    if (t == T_NUM) {
+      uint64_t b = (uint64_t)__double_as_longlong(n);
+      soff[f * BLOCK + tid] = (uint32_t)b;
+      slen[f * BLOCK + tid] = (uint32_t)(b >> 32);
+    } else {
+      soff[f * BLOCK + tid] = off;
+      slen[f * BLOCK + tid] = len;
+    }
   }
   DEV void set_len(int f, uint32_t len) { slen[f * BLOCK + tid] = len; }
   DEV uint8_t  get_type(int f) const { return type[f * BLOCK + tid]; }
   DEV uint32_t get_soff(int f) const { return soff[f * BLOCK + tid]; }
   DEV uint32_t get_slen(int f) const { return slen[f * BLOCK + tid]; }
-  DEV double   get_num(int f) const { return num[f * BLOCK + tid]; }
+  DEV double   get_num(int f) const {
+    uint64_t b = (uint64_t)soff[f * BLOCK + tid]
+               | ((uint64_t)slen[f * BLOCK + tid] << 32);
+    return __longlong_as_double((long long)b);
+  }
 };
 
 // -------------------------------------------------------------------
@@ -1567,8 +1580,6 @@ __global__ void columnar_query_kernel(ColArgs A) {
   const PlanView& P = A.P;
   const int nf = P.nf;
   size_t off = 0;
-  double* fv_num = reinterpret_cast<double*>(smemc + off);
-  off += (size_t)nf * BLOCK * sizeof(double);
   uint32_t* fv_soff = reinterpret_cast<uint32_t*>(smemc + off);
   off += (size_t)nf * BLOCK * sizeof(uint32_t);
   uint32_t* fv_slen = reinterpret_cast<uint32_t*>(smemc + off);
@@ -1593,7 +1604,7 @@ __global__ void columnar_query_kernel(ColArgs A) {
 
   FV fv;
   fv.type = fv_type; fv.soff = fv_soff; fv.slen = fv_slen;
-  fv.num = fv_num; fv.tid = threadIdx.x;
+  fv.tid = threadIdx.x;
   Bytes BV;
   BV.mem = A.blob;
   BV.bias = 0;
@@ -2088,8 +2099,6 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
 
   // LDS layout: fv SoA | agg cache | counters
   size_t off = 0;
-  double* fv_num = reinterpret_cast<double*>(smem + off);
-  off += (size_t)nf * BLOCK * sizeof(double);
   uint32_t* fv_soff = reinterpret_cast<uint32_t*>(smem + off);
   off += (size_t)nf * BLOCK * sizeof(uint32_t);
   uint32_t* fv_slen = reinterpret_cast<uint32_t*>(smem + off);
@@ -2123,7 +2132,7 @@ DEV void scan_kernel_body(char* smem, ScanArgs A) {
   __syncthreads();
 
   FV fv;
-  fv.type = fv_type; fv.soff = fv_soff; fv.slen = fv_slen; fv.num = fv_num;
+  fv.type = fv_type; fv.soff = fv_soff; fv.slen = fv_slen;
   fv.tid = threadIdx.x;
 
   if constexpr (XP != 0) {

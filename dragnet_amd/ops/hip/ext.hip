@@ -237,6 +237,9 @@ void scan_chunk(
     else if (mw == 3)
       hipLaunchKernelGGL((scan_kernel_x<3, 6>), dim3(blocks),
                          dim3(BLOCK), lds, current_stream(), A);
+    else if (mw == 5)
+      hipLaunchKernelGGL((scan_kernel_x<5, 6>), dim3(blocks),
+                         dim3(BLOCK), lds, current_stream(), A);
     else
       hipLaunchKernelGGL((scan_kernel_x<4, 6>), dim3(blocks),
                          dim3(BLOCK), lds, current_stream(), A);

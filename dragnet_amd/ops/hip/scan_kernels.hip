@@ -24,7 +24,10 @@ namespace dn {
 
 constexpr int BLOCK = 256;
 constexpr int SIG_DEPTH = 6;      // max dotted-path components
-constexpr int LDS_CACHE = 256;    // per-block aggregation cache slots
+constexpr int LDS_CACHE = 128;    // per-block aggregation cache slots
+// (halved from 256 in r2: the 7 KB buys a whole extra block of
+// occupancy, which outweighs extra cache-miss fallthrough; misses
+// go to the dense partial row / global table either way)
 
 // -------------------------------------------------------------------
 // small utilities
@@ -1825,6 +1828,7 @@ __global__ void scan_kernel_x(ScanArgs A) {
 template __global__ void scan_kernel_x<2, 6>(ScanArgs);
 template __global__ void scan_kernel_x<3, 6>(ScanArgs);
 template __global__ void scan_kernel_x<4, 6>(ScanArgs);
+template __global__ void scan_kernel_x<5, 6>(ScanArgs);
 template __global__ void scan_kernel_x<4, 5>(ScanArgs);
 template __global__ void scan_kernel_x<4, 7>(ScanArgs);
 

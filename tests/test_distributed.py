@@ -378,3 +378,54 @@ def test_world4_empty_shard_and_empty_tables(fixture_tree):
     for _rank, p1, p2 in results:
         assert p1 == expected
         assert p2 == []  # empty table merges to empty everywhere
+
+
+def _worker_floatw(rank, world, port, out_q):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    from dragnet_amd.distributed import (init_process_group,
+                                         merge_tables_tensor)
+    from dragnet_amd.points import Aggregator
+    from dragnet_amd.query import query_load
+    init_process_group(backend="gloo")
+    q = query_load(breakdown_specs="k,lat[aggr=quantize]")
+    agg = Aggregator(q)
+    # float weights (json-skinner reduce shape), negative ordinals,
+    # a shared key and rank-unique keys
+    agg.table[("shared", 3)] = 0.5 + rank        # 0.5 / 1.5
+    agg.table[("r%d" % rank, -2)] = 1.25
+    agg.table[("int", 1)] = 2
+    merged = merge_tables_tensor(agg, q)
+    out_q.put((rank, dict(merged.table)))
+    import torch.distributed as dist
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_merge_float_weights_and_negative_ordinals():
+    """Dense merge wire format: non-integer f64 values survive
+    exactly, negative ordinals round-trip, int-typed totals stay
+    ints."""
+    ctx = mp.get_context("spawn")
+    out_q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_floatw, args=(r, 2, 29541,
+                                                      out_q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [out_q.get(timeout=90) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    expected = {("shared", 3): 2.0, ("r0", -2): 1.25,
+                ("r1", -2): 1.25, ("int", 1): 4}
+    for _rank, table in results:
+        assert table == expected
+        assert isinstance(table[("int", 1)], int)
+        assert isinstance(table[("r0", -2)], float)

@@ -1,0 +1,75 @@
+"""Kernel band across query shapes on the FINAL r2 kernels — linear
+and wave-transposed resident scans (updates the r1 165-245 GB/s
+claim)."""
+import os
+import sys
+import time
+
+sys.path.insert(0, "/root/repo")
+
+import torch  # noqa: E402
+
+from dragnet_amd.engine import plan as planmod  # noqa: E402
+from dragnet_amd.engine.gpu import GpuEngine, _ScanContext  # noqa: E402
+from dragnet_amd.query import query_load  # noqa: E402
+from dragnet_amd.tools.mktestdata import generate_lines  # noqa: E402
+
+lines = []
+total = 0
+for line in generate_lines(1 << 62, seed=9):
+    lines.append(line)
+    total += len(line)
+    if total >= 256 << 20:
+        break
+pool = b"".join(lines)
+nrec = len(lines)
+eng = GpuEngine()
+eng.chunk_bytes = len(pool)
+
+CASES = [
+    ("count-only", query_load()),
+    ("flagship filter+2bd", query_load(
+        filter={"eq": ["req.method", "GET"]},
+        breakdown_specs="req.method,res.statusCode")),
+    ("pure-ordinal lquantize", query_load(
+        breakdown_specs="dataLatency[aggr=lquantize,step=100]")),
+    ("1 string bd", query_load(breakdown_specs="req.method")),
+    ("high-card string (urls)", query_load(breakdown_specs="req.url")),
+    ("5-field bd", query_load(
+        breakdown_specs="host,operation,req.method,res.statusCode,"
+                        "latency[aggr=quantize]")),
+    ("filter-only count", query_load(
+        filter={"eq": ["req.method", "GET"]})),
+    ("date lquantize (synth)", query_load(
+        breakdown_specs="t[date,field=time,aggr=lquantize,step=60]")),
+]
+
+for mode in ("linear", "xpose"):
+    print("== %s resident ==" % mode)
+    for name, q in CASES:
+        cplan = planmod.compile_plan([q])
+        ctx = _ScanContext(eng, cplan, 1 << 18, 1 << 18, 32 << 20)
+        if mode == "xpose":
+            ctx.stage_xpose(pool)
+
+            def one(c=ctx):
+                c.reset()
+                c.scan_xpose()
+        else:
+            ctx.stage_resident(pool)
+
+            def one(c=ctx):
+                c.reset()
+                c.scan_resident(h2d=False)
+        for _ in range(2):
+            one()
+        torch.cuda.synchronize()
+        t0 = time.time()
+        for _ in range(6):
+            one()
+        torch.cuda.synchronize()
+        dt = (time.time() - t0) / 6
+        print("%-28s %7.1f GB/s %8.1f M rec/s"
+              % (name, len(pool) / dt / 1e9, nrec / dt / 1e6),
+              flush=True)
+        del ctx

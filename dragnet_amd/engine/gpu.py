@@ -250,7 +250,7 @@ class _ScanContext(object):
             dense = _env_int("DRAGNET_DENSE", 1) == 1
         self.dense = dense
         if dense:
-            agg_slots = _env_int("DRAGNET_DENSE_SLOTS", 4096)
+            agg_slots = _env_int("DRAGNET_DENSE_SLOTS", 8192)
         i32 = dict(dtype=torch.int32, device=dev)
         f64 = dict(dtype=torch.float64, device=dev)
         u8 = dict(dtype=torch.uint8, device=dev)

@@ -90,7 +90,6 @@ def main():
         log("self-launching %d ranks via torch.distributed.run"
             % args.gpus)
         sys.exit(subprocess.call(cmd))
-    n_gpus = max(args.gpus, world)
 
     import torch
     assert torch.cuda.is_available(), "bench needs a GPU"

@@ -737,6 +737,10 @@ class _ScanContext(object):
         torch = self.t
         dev = self.eng.device
         gran = _env_int("DRAGNET_XGRAN", 64)
+        if gran not in (32, 64, 128):
+            # the scan kernel is templated on the granule size; an
+            # unsupported value would mismatch the staged layout
+            raise ValueError("DRAGNET_XGRAN must be 32, 64 or 128")
         glog = gran.bit_length() - 1
         if not hasattr(self, "_resident"):
             if buf is None:
@@ -787,6 +791,8 @@ class _ScanContext(object):
         torch = self.t
         dev = self.eng.device
         gran = _env_int("DRAGNET_XGRAN", 64)
+        if gran not in (32, 64, 128):
+            raise ValueError("DRAGNET_XGRAN must be 32, 64 or 128")
         xb, wave_base, slot_len, nslots, n = _build_xpose_layout(
             buf, gran)
         self._x = {

@@ -1,0 +1,99 @@
+"""Property-based invariants of the CPU oracle (hypothesis): for ANY
+byte input and any well-formed query, every pipeline stage conserves
+records (ninputs == noutputs + attributed drops), stage chaining is
+exact, and the aggregate total equals the accepted record count.
+These are the same invariants the premature-exit guard enforces at
+runtime and the GPU engine is differential-tested against."""
+
+import json
+
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from dragnet_amd.query import query_load
+from dragnet_amd.scan_cpu import ScanPipeline
+
+scalars = st.one_of(
+    st.none(), st.booleans(),
+    st.integers(min_value=-10**12, max_value=10**12),
+    st.floats(allow_nan=False, allow_infinity=False, width=32),
+    st.text(max_size=12))
+
+values = st.recursive(
+    scalars,
+    lambda children: st.one_of(
+        st.lists(children, max_size=3),
+        st.dictionaries(st.text(max_size=6), children, max_size=4)),
+    max_leaves=8)
+
+records = st.dictionaries(
+    st.sampled_from(["a", "b", "req", "time", "lat", "x.y"]),
+    values, max_size=5)
+
+lines = st.lists(st.one_of(
+    records.map(lambda r: json.dumps(r).encode()),
+    st.binary(max_size=30).map(lambda b: b.replace(b"\n", b"x")),
+    st.just(b""),
+), max_size=40)
+
+queries = st.sampled_from([
+    {},
+    {"breakdown_specs": "a"},
+    {"breakdown_specs": "req.m,b"},
+    {"breakdown_specs": "lat[aggr=quantize]"},
+    {"filter": {"eq": ["a", 1]}},
+    {"filter": {"and": [{"gt": ["lat", 0]}, {"ne": ["b", "x"]}]},
+     "breakdown_specs": "b"},
+    {"breakdown_specs": "t[date,field=time,aggr=lquantize,step=60]"},
+])
+
+DROPS = ("invalid json", "nfilteredout", "nfailedeval", "undef",
+         "baddate", "nonnumeric")
+
+
+@settings(max_examples=120, deadline=None)
+@given(lines=lines, qkw=queries)
+def test_pipeline_conservation(lines, qkw):
+    q = query_load(**qkw)
+    p = ScanPipeline(q)
+    p.write_bytes(b"".join(ln + b"\n" for ln in lines))
+    p.finish()
+    stages = p.counter_stages()
+
+    prev_out = None
+    for name, c in stages:
+        if name == "Aggregator":
+            continue
+        drops = sum(c.get(k, 0) for k in DROPS)
+        assert c["ninputs"] == c["noutputs"] + drops, (name, c)
+        if prev_out is not None:
+            assert c["ninputs"] == prev_out, (name, c)
+        prev_out = c["noutputs"]
+
+    agg = dict(stages)["Aggregator"]
+    assert agg["ninputs"] == prev_out
+    accepted = agg["ninputs"] - agg.get("nonnumeric", 0)
+    total = sum(p.aggr.table.values()) if p.aggr.table else 0
+    assert total == accepted
+
+
+@settings(max_examples=60, deadline=None)
+@given(lines=lines)
+def test_point_reaggregation_idempotent(lines):
+    """points() -> re-aggregate x3 triples values exactly (the
+    load-bearing associativity the distributed merge rests on)."""
+    q = query_load(breakdown_specs="a,b")
+    p = ScanPipeline(q)
+    p.write_bytes(b"".join(ln + b"\n" for ln in lines))
+    p.finish()
+    pts = p.aggr.points()
+
+    from dragnet_amd.points import Aggregator
+    agg3 = Aggregator(q)
+    for _ in range(3):
+        for pt in pts:
+            agg3.write({"fields": dict(pt["fields"]),
+                        "value": pt["value"]})
+    assert agg3.points() == [
+        {"fields": dict(pt["fields"]), "value": 3 * pt["value"]}
+        for pt in pts]

@@ -97,3 +97,33 @@ def test_point_reaggregation_idempotent(lines):
     assert agg3.points() == [
         {"fields": dict(pt["fields"]), "value": 3 * pt["value"]}
         for pt in pts]
+
+
+@settings(max_examples=100, deadline=None)
+@given(items=st.lists(
+    st.tuples(
+        st.tuples(
+            st.one_of(st.text(max_size=8),
+                      st.integers(min_value=-2**40, max_value=2**40)),
+            st.one_of(st.text(max_size=4),
+                      st.integers(min_value=-100, max_value=100))),
+        st.one_of(st.integers(min_value=1, max_value=10**9),
+                  st.floats(min_value=0.001, max_value=1e9,
+                            allow_nan=False))),
+    max_size=30, unique_by=lambda kv: kv[0]))
+def test_merge_wire_format_roundtrip(items):
+    """The dense-merge encoder/decoder round-trips ANY table exactly
+    (str/int key elements, int/float values) — the single-rank slice
+    of the C1 wire format."""
+    from dragnet_amd.distributed import _encode_table, _rebuild_table
+    from dragnet_amd.points import Aggregator
+    from dragnet_amd.query import query_load
+    q = query_load(breakdown_specs="a,b")
+    agg = Aggregator(q)
+    for k, v in items:
+        agg.table[k] = v
+    codes, tags, vals, strings = _encode_table(agg, 2)
+    out = _rebuild_table(q, codes, tags, vals, strings)
+    want = {k: (int(v) if float(v).is_integer() else float(v))
+            for k, v in agg.table.items()}
+    assert out.table == want

@@ -160,7 +160,6 @@ struct ScanArgs {
   NumDict ndict;
   unsigned long long* counters;
   int data_format_skinner;
-  int wave_intern;  // wave-deduplicated dictionary interning
   // wave-transposed staging (scan_kernel_x; null for linear scans)
   const uint8_t* xdata;              // transposed pool
   const unsigned long long* xwave_base;  // [nwaves] byte base per wave

@@ -176,8 +176,6 @@ void scan_chunk(
   A.ndict = make_ndict(nd_state, nd_bits, nd_id, nd_next);
   A.counters = (unsigned long long*)counters.data_ptr();
   A.data_format_skinner = skinner ? 1 : 0;
-  const char* wi_env = getenv("DRAGNET_WAVE_INTERN");
-  A.wave_intern = (wi_env && atoi(wi_env)) ? 1 : 0;
   A.xdata = nullptr; A.xwave_base = nullptr; A.xrec_len = nullptr;
   A.xn_slots = 0;
   if (xn_slots > 0) {

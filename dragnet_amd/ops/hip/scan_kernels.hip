@@ -1186,8 +1186,9 @@ DEV uint64_t hash_bytes(BS BV, uint32_t off, uint32_t len) {
 
 // Returns string id, or 0xFFFFFFFF on table/data overflow.
 template <class BS>
-DEV uint32_t intern_string_h(const StrDict& D, BS BV,
-                             uint32_t off, uint32_t len, uint64_t h) {
+DEV uint32_t intern_string(const StrDict& D, BS BV,
+                           uint32_t off, uint32_t len) {
+  uint64_t h = hash_bytes(BV, off, len);
   uint32_t mask = D.nslots - 1;
   uint32_t s = (uint32_t)h & mask;
   for (uint32_t probes = 0; probes < D.nslots; probes++, s = (s + 1) & mask) {
@@ -1232,43 +1233,10 @@ DEV uint32_t intern_string_h(const StrDict& D, BS BV,
   return 0xFFFFFFFFu;
 }
 
-template <class BS>
-DEV uint32_t intern_string(const StrDict& D, BS BV,
-                           uint32_t off, uint32_t len) {
-  return intern_string_h(D, BV, off, len, hash_bytes(BV, off, len));
-}
-
-// Wave-deduplicated intern: lanes carrying the same span hash elect a
-// leader which does the dictionary probe (and its per-byte verify);
-// the id broadcasts back.  One probe per DISTINCT value per wave
-// instead of per record — the win on low-cardinality group-by
-// columns; a fully-distinct wave degrades to a 64-iteration shuffle
-// loop, so the path is gated (ScanArgs.wave_intern).  Two DIFFERENT
-// strings sharing a 64-bit span hash would share an id (~2^-64 per
-// pair; same accepted corner as the LDS cache kh).
-template <class BS>
-DEV uint32_t intern_string_wave(const StrDict& D, BS BV,
-                                uint32_t off, uint32_t len) {
-  uint64_t h = hash_bytes(BV, off, len);
-  uint64_t unproc = __ballot(true);
-  uint32_t out = 0xFFFFFFFFu;
-  while (unproc) {
-    int src = (int)(__ffsll((long long)unproc) - 1);
-    uint64_t src_h = __shfl(h, src);
-    bool same = (h == src_h);
-    uint64_t grp = __ballot(same);
-    uint32_t id = 0;
-    if ((int)__lane_id() == src)
-      id = intern_string_h(D, BV, off, len, h);
-    id = __shfl(id, src);
-    if (same) out = id;
-    unproc &= ~grp;
-  }
-  return out;
-}
-
-DEV uint32_t intern_number_h(const NumDict& D, uint64_t bits,
-                             uint64_t h) {
+DEV uint32_t intern_number(const NumDict& D, double v) {
+  if (v == 0.0) v = 0.0;  // canonicalize -0
+  uint64_t bits = __double_as_longlong(v);
+  uint64_t h = mix64(bits ^ 0x9E3779B97F4A7C15ull);
   uint32_t mask = D.nslots - 1;
   uint32_t s = (uint32_t)h & mask;
   for (uint32_t probes = 0; probes < D.nslots; probes++, s = (s + 1) & mask) {
@@ -1294,35 +1262,6 @@ DEV uint32_t intern_number_h(const NumDict& D, uint64_t bits,
     }
   }
   return 0xFFFFFFFFu;
-}
-
-DEV uint32_t intern_number(const NumDict& D, double v) {
-  if (v == 0.0) v = 0.0;  // canonicalize -0
-  uint64_t bits = __double_as_longlong(v);
-  return intern_number_h(D, bits, mix64(bits ^ 0x9E3779B97F4A7C15ull));
-}
-
-// Wave-deduplicated number intern (exact: dedupe is on the bit
-// pattern itself).
-DEV uint32_t intern_number_wave(const NumDict& D, double v) {
-  if (v == 0.0) v = 0.0;
-  uint64_t bits = __double_as_longlong(v);
-  uint64_t unproc = __ballot(true);
-  uint32_t out = 0xFFFFFFFFu;
-  while (unproc) {
-    int src = (int)(__ffsll((long long)unproc) - 1);
-    uint64_t src_b = __shfl(bits, src);
-    bool same = (bits == src_b);
-    uint64_t grp = __ballot(same);
-    uint32_t id = 0;
-    if ((int)__lane_id() == src)
-      id = intern_number_h(D, bits,
-                           mix64(bits ^ 0x9E3779B97F4A7C15ull));
-    id = __shfl(id, src);
-    if (same) out = id;
-    unproc &= ~grp;
-  }
-  return out;
 }
 
 // -------------------------------------------------------------------
@@ -2082,15 +2021,11 @@ DEV void process_record(BS BV, uint32_t start, uint32_t end,
             }
             code = make_code(TAG_SPECIAL, SPECIAL_ARRJSON | (id << 3));
           } else if (t == T_NUM) {
-            uint32_t id = A.wave_intern
-                              ? intern_number_wave(A.ndict, num)
-                              : intern_number(A.ndict, num);
+            uint32_t id = intern_number(A.ndict, num);
             if (id == 0xFFFFFFFFu) { overflow = true; break; }
             code = make_code(TAG_NUM, id);
           } else {  // T_STR
-            uint32_t id = A.wave_intern
-                              ? intern_string_wave(A.sdict, BV, so, sl)
-                              : intern_string(A.sdict, BV, so, sl);
+            uint32_t id = intern_string(A.sdict, BV, so, sl);
             if (id == 0xFFFFFFFFu) { overflow = true; break; }
             code = make_code(TAG_STR, id);
           }

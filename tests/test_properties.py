@@ -127,3 +127,62 @@ def test_merge_wire_format_roundtrip(items):
     want = {k: (int(v) if float(v).is_integer() else float(v))
             for k, v in agg.table.items()}
     assert out.table == want
+
+
+@settings(max_examples=150, deadline=None)
+@given(y=st.integers(min_value=1970, max_value=2199),
+       mo=st.integers(min_value=1, max_value=12),
+       d=st.integers(min_value=1, max_value=31),
+       hh=st.integers(min_value=0, max_value=23),
+       mm=st.integers(min_value=0, max_value=59),
+       ss=st.integers(min_value=0, max_value=59),
+       ms=st.integers(min_value=0, max_value=999))
+def test_jsdate_matches_datetime(y, mo, d, hh, mm, ss, ms):
+    """parse_ms agrees with Python's datetime for every valid
+    Z-suffixed timestamp; invalid day-of-month yields None."""
+    import datetime as dt
+
+    from dragnet_amd import jsdate
+    s = "%04d-%02d-%02dT%02d:%02d:%02d.%03dZ" % (y, mo, d, hh, mm,
+                                                 ss, ms)
+    try:
+        t = dt.datetime(y, mo, d, hh, mm, ss, ms * 1000,
+                        tzinfo=dt.timezone.utc)
+        want = int(t.timestamp() * 1000)
+    except ValueError:
+        want = None
+    assert jsdate.parse_ms(s) == want
+    if want is not None:
+        # round trip through to_iso
+        assert jsdate.to_iso(want / 1000.0) == s
+
+
+@settings(max_examples=150, deadline=None)
+@given(parts=st.lists(
+    st.tuples(
+        st.text(alphabet="abcxyz_.", min_size=1, max_size=8)
+          .filter(lambda t: "," not in t and "[" not in t),
+        st.sampled_from([None, "date", "quantize"])),
+    min_size=1, max_size=4))
+def test_attrs_roundtrip(parts):
+    """attrs_parse on a spec rendered from structured parts recovers
+    the same names/attrs (modulo the documented reference off-by-one
+    we deliberately FIX)."""
+    from dragnet_amd.attrs import attrs_parse
+    frags = []
+    for name, attr in parts:
+        if attr == "date":
+            frags.append("%s[date]" % name)
+        elif attr == "quantize":
+            frags.append("%s[aggr=quantize]" % name)
+        else:
+            frags.append(name)
+    spec = ",".join(frags)
+    out = attrs_parse(spec)
+    assert not isinstance(out, Exception), (spec, out)
+    assert [o["name"] for o in out] == [p[0] for p in parts]
+    for o, (name, attr) in zip(out, parts):
+        if attr == "date":
+            assert "date" in o
+        elif attr == "quantize":
+            assert o.get("aggr") == "quantize"

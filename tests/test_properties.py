@@ -146,9 +146,11 @@ def test_jsdate_matches_datetime(y, mo, d, hh, mm, ss, ms):
     s = "%04d-%02d-%02dT%02d:%02d:%02d.%03dZ" % (y, mo, d, hh, mm,
                                                  ss, ms)
     try:
-        t = dt.datetime(y, mo, d, hh, mm, ss, ms * 1000,
+        t = dt.datetime(y, mo, d, hh, mm, ss,
                         tzinfo=dt.timezone.utc)
-        want = int(t.timestamp() * 1000)
+        # integer epoch math: float timestamp()*1000 truncates the
+        # millisecond (this very test caught that in its first form)
+        want = int(t.timestamp()) * 1000 + ms
     except ValueError:
         want = None
     assert jsdate.parse_ms(s) == want

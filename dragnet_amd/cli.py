@@ -725,9 +725,17 @@ def cmd_index_read(argv):
     from .datasource.file import metric_query, write_index
     queries = [metric_query(m, interval, "__dn_ts")
                for m in metrics]
-    # Re-aggregate incoming tagged points per metric.
+    # Re-aggregate incoming tagged points per metric: the C++ reducer
+    # (index/_points) takes the flat-scalar fast shape and returns the
+    # rest for the Python oracle below (DRAGNET_PY_POINTS=1 forces the
+    # all-Python path).
     aggs = [Aggregator(q) for q in queries]
-    for line in sys.stdin.buffer:
+    from .points import reduce_tagged_stream
+    try:
+        lines = reduce_tagged_stream(sys.stdin.buffer, aggs, queries)
+    except ImportError:
+        lines = sys.stdin.buffer
+    for line in lines:
         line = line.strip()
         if not line:
             continue

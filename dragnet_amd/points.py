@@ -196,3 +196,61 @@ class Flattener(object):
 
     def rows(self):
         return self.agg.rows()
+
+
+def fast_reduce_spec(query):
+    """Column spec for the C++ tagged-point reducer (_points): one
+    (name, kind, step) per breakdown, kinds 0=canonical 1=date
+    2=quantize 3=lquantize — mirrors Aggregator.write exactly."""
+    cols = []
+    for b in query.breakdowns:
+        name = b["name"]
+        bk = query.bucketizers.get(name)
+        if bk is None:
+            cols.append((name, 1 if "date" in b else 0, 0.0))
+        elif bk.aggr == "quantize":
+            cols.append((name, 2, 0.0))
+        else:
+            cols.append((name, 3, float(bk.step)))
+    return cols
+
+
+def reduce_tagged_stream(stream, aggs, queries):
+    """Drain a tagged-point NDJSON stream through the C++ reducer,
+    merging fast-path groups into each Aggregator and returning the
+    punted lines (anything outside the flat-scalar fast shape) for
+    the caller's per-line Python path.  Raises ImportError when the
+    native extension is unavailable (caller falls back wholesale)."""
+    import os
+    if os.environ.get("DRAGNET_PY_POINTS") == "1":
+        raise ImportError("DRAGNET_PY_POINTS=1")
+    from .index import _points
+
+    specs = [fast_reduce_spec(q) for q in queries]
+    punted_all = []
+
+    def run(data):
+        tables, nin, nonn, punted = _points.reduce_tagged(data, specs)
+        for agg, t, n, d in zip(aggs, tables, nin, nonn):
+            agg.ninputs += n
+            agg.ndropped_nonnumeric += d
+            tab = agg.table
+            for k, v in t.items():
+                tab[k] = tab.get(k, 0) + v
+        punted_all.extend(punted)
+
+    rem = b""
+    while True:
+        chunk = stream.read(64 << 20)
+        if not chunk:
+            break
+        chunk = rem + chunk if rem else chunk
+        cut = chunk.rfind(b"\n")
+        if cut < 0:
+            rem = chunk
+            continue
+        rem = chunk[cut + 1:]
+        run(chunk[:cut + 1])
+    if rem:
+        run(rem)
+    return punted_all

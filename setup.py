@@ -46,6 +46,14 @@ setup(
             },
         ),
         Extension(
+            name="dragnet_amd.index._points",
+            sources=["dragnet_amd/index/points_fast.cpp"],
+            include_dirs=[pybind11.get_include()],
+            extra_compile_args=["-O3", "-std=c++17",
+                                "-fvisibility=hidden"],
+            language="c++",
+        ),
+        Extension(
             name="dragnet_amd.index._csink",
             sources=["dragnet_amd/index/csink.cpp"],
             include_dirs=[pybind11.get_include()],

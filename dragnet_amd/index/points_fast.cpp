@@ -582,10 +582,150 @@ py::tuple reduce_tagged(
   return py::make_tuple(tables, nin, nonn, plines);
 }
 
+// ---- emit side: byte-exact json.dumps(separators=(",",":")) ----
+// for output_points (dn scan --points / index-scan map emit).  Any
+// value outside str/int/float/bool/None falls back to the Python
+// point_json callable for THAT point, preserving output order.
+
+void esc_json(std::string& out, PyObject* u) {
+  // matches c_encode_basestring_ascii: ensure_ascii, lowercase hex
+  static const char* hexd = "0123456789abcdef";
+  Py_ssize_t n = PyUnicode_GET_LENGTH(u);
+  int kind = PyUnicode_KIND(u);
+  const void* dat = PyUnicode_DATA(u);
+  out.push_back('"');
+  for (Py_ssize_t i = 0; i < n; ++i) {
+    Py_UCS4 cp = PyUnicode_READ(kind, dat, i);
+    if (cp == '"') { out += "\\\""; continue; }
+    if (cp == '\\') { out += "\\\\"; continue; }
+    if (cp >= 0x20 && cp < 0x7F) {
+      out.push_back(static_cast<char>(cp));
+      continue;
+    }
+    switch (cp) {
+      case 0x08: out += "\\b"; continue;
+      case 0x09: out += "\\t"; continue;
+      case 0x0A: out += "\\n"; continue;
+      case 0x0C: out += "\\f"; continue;
+      case 0x0D: out += "\\r"; continue;
+    }
+    auto u4 = [&](uint32_t v) {
+      out += "\\u";
+      out.push_back(hexd[(v >> 12) & 15]);
+      out.push_back(hexd[(v >> 8) & 15]);
+      out.push_back(hexd[(v >> 4) & 15]);
+      out.push_back(hexd[v & 15]);
+    };
+    if (cp >= 0x10000) {
+      uint32_t v = cp - 0x10000;
+      u4(0xD800 + (v >> 10));
+      u4(0xDC00 + (v & 0x3FF));
+    } else {
+      u4(cp);
+    }
+  }
+  out.push_back('"');
+}
+
+bool emit_scalar(std::string& out, PyObject* o) {
+  if (o == Py_None) { out += "null"; return true; }
+  if (PyBool_Check(o)) {
+    out += (o == Py_True) ? "true" : "false";
+    return true;
+  }
+  if (PyLong_Check(o)) {
+    int overflow = 0;
+    long long v = PyLong_AsLongLongAndOverflow(o, &overflow);
+    if (!overflow) {
+      char buf[24];
+      out.append(buf, static_cast<size_t>(
+          snprintf(buf, sizeof(buf), "%lld", v)));
+      return true;
+    }
+    PyObject* r = PyObject_Str(o);  // big ints: int.__repr__, ascii
+    if (r == nullptr) throw py::error_already_set();
+    Py_ssize_t len;
+    const char* c = PyUnicode_AsUTF8AndSize(r, &len);
+    out.append(c, static_cast<size_t>(len));
+    Py_DECREF(r);
+    return true;
+  }
+  if (PyFloat_Check(o)) {
+    double d = PyFloat_AS_DOUBLE(o);
+    if (std::isnan(d)) { out += "NaN"; return true; }
+    if (std::isinf(d)) {
+      out += d > 0 ? "Infinity" : "-Infinity";
+      return true;
+    }
+    PyObject* r = PyObject_Repr(o);  // json uses float.__repr__
+    if (r == nullptr) throw py::error_already_set();
+    Py_ssize_t len;
+    const char* c = PyUnicode_AsUTF8AndSize(r, &len);
+    out.append(c, static_cast<size_t>(len));
+    Py_DECREF(r);
+    return true;
+  }
+  if (PyUnicode_Check(o)) {
+    esc_json(out, o);
+    return true;
+  }
+  return false;  // nested/list/other: fall back to point_json
+}
+
+// serialize_points(points, fallback) -> bytes (one NDJSON line each;
+// fallback(point) -> str handles points with non-scalar values)
+py::bytes serialize_points(py::sequence points, py::object fallback) {
+  std::string out;
+  out.reserve(1 << 16);
+  for (py::handle ph : points) {
+    size_t mark = out.size();
+    PyObject* p = ph.ptr();
+    PyObject* fields = PyDict_Check(p)
+                           ? PyDict_GetItemString(p, "fields")
+                           : nullptr;
+    PyObject* value = PyDict_Check(p)
+                          ? PyDict_GetItemString(p, "value")
+                          : nullptr;
+    bool ok = fields != nullptr && PyDict_Check(fields) &&
+              value != nullptr;
+    if (ok) {
+      out += "{\"fields\":{";
+      PyObject *k, *v;
+      Py_ssize_t pos = 0;
+      bool first = true;
+      while (ok && PyDict_Next(fields, &pos, &k, &v)) {
+        if (!PyUnicode_Check(k)) { ok = false; break; }
+        if (!first) out.push_back(',');
+        first = false;
+        esc_json(out, k);
+        out.push_back(':');
+        ok = emit_scalar(out, v);
+      }
+      if (ok) {
+        out += "},\"value\":";
+        ok = emit_scalar(out, value);
+        if (ok) out += "}\n";
+      }
+    }
+    if (!ok) {
+      out.resize(mark);  // rewind partial line, use the Python path
+      py::str line = fallback(ph);
+      Py_ssize_t len;
+      const char* c = PyUnicode_AsUTF8AndSize(line.ptr(), &len);
+      if (c == nullptr) throw py::error_already_set();
+      out.append(c, static_cast<size_t>(len));
+      out.push_back('\n');
+    }
+  }
+  return py::bytes(out);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(_points, m) {
   m.doc() = "dragnet_amd fast tagged-point reducer (C++)";
   m.def("reduce_tagged", &reduce_tagged, py::arg("data"),
         py::arg("specs"));
+  m.def("serialize_points", &serialize_points, py::arg("points"),
+        py::arg("fallback"));
 }

@@ -168,7 +168,26 @@ def point_json(point):
 
 
 def output_points(points, out=None):
+    """Emit points as NDJSON — through the C++ serializer
+    (index/_points, byte-exact json.dumps; any point with a
+    non-scalar value falls back to point_json in place) unless
+    DRAGNET_PY_POINTS=1 or the extension is absent."""
+    import os
     out = out or sys.stdout
+    if os.environ.get("DRAGNET_PY_POINTS") != "1":
+        try:
+            from .index import _points
+        except ImportError:
+            pass
+        else:
+            data = _points.serialize_points(list(points), point_json)
+            buf = getattr(out, "buffer", None)
+            if buf is not None:
+                out.flush()  # keep text-layer writes ordered
+                buf.write(data)
+            else:
+                out.write(data.decode("utf-8"))
+            return
     for p in points:
         out.write(point_json(p) + "\n")
 

@@ -190,3 +190,52 @@ def test_streaming_slab_boundary():
     ref = py_reduce(data, QUERIES)
     assert aggs[0].table == ref[0].table
     assert aggs[0].ninputs == ref[0].ninputs
+
+
+def test_serializer_byte_exact():
+    """serialize_points output is byte-identical to per-point
+    json.dumps across escapes, unicode, numbers, and fallback
+    shapes (nested values)."""
+    from dragnet_amd.index import _points
+    from dragnet_amd.output import point_json
+    pts = [
+        {"fields": {"a": "x", "n": 200, "f": 2.5}, "value": 1},
+        {"fields": {"a": 'q"\\\n\t\x07\x7f', "b": "café",
+                    "c": "\U0001F600"}, "value": 3},
+        {"fields": {"neg": -17, "big": 10**25, "z": -0.0,
+                    "e": 1e-5, "E": 1e16}, "value": 0.25},
+        {"fields": {"t": True, "f2": False, "n2": None}, "value": 7},
+        {"fields": {}, "value": 2},
+        {"fields": {"nested": {"x": 1}, "l": [1, "a", None]},
+         "value": 1},  # falls back to point_json
+        {"fields": {"inf": float("inf"), "nan": float("nan")},
+         "value": 1},
+    ]
+    got = _points.serialize_points(pts, point_json)
+    want = b"".join(point_json(p).encode() + b"\n" for p in pts)
+    assert got == want
+
+
+def test_serializer_reducer_roundtrip():
+    """points -> serialize -> reduce recovers the exact table (the
+    map/reduce pipe identity the index build rests on)."""
+    from dragnet_amd.index import _points
+    from dragnet_amd.output import point_json
+    q = QUERIES[0]
+    src = Aggregator(q)
+    rng = random.Random(11)
+    for i in range(500):
+        src.write({"fields": {"a": "k%d" % (i % 17),
+                              "t": 1400000000 + i,
+                              "lat": rng.choice([1, 5, 80]),
+                              "ts": rng.randrange(0, 600)},
+                   "value": rng.randint(1, 9)})
+    pts = src.points()
+    for p in pts:
+        p["fields"]["__dn_metric"] = 0
+    data = _points.serialize_points(pts, point_json)
+    aggs = [Aggregator(qq) for qq in QUERIES]
+    punted = reduce_tagged_stream(io.BytesIO(data), aggs, QUERIES)
+    assert punted == []
+    assert aggs[0].table == src.table
+    assert sum(aggs[0].table.values()) == sum(src.table.values())

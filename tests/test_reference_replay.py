@@ -557,3 +557,162 @@ def test_scan_manta_goldens_sharded(tmp_path):
     # rank 1 prints nothing
     assert all(code == 0 and out == ""
                for code, out in results[1])
+
+
+def _worker_index_manta(rank, world, port, cfgfile, idx_root,
+                        arglists, out_q):
+    """2-rank sharded replay of tst.index_manta.sh: distributed
+    builds interleaved with rank-0 config mutations (barriers keep
+    the phases in lockstep), queries on every rank (rank 1 prints
+    nothing by design)."""
+    import io
+    import shutil
+    import sys as _sys
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["DRAGNET_CONFIG"] = cfgfile
+    os.environ["DRAGNET_ENGINE"] = "cpu"
+    _sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    from dragnet_amd import cli
+    from dragnet_amd.distributed import init_process_group
+    dist = init_process_group(backend="gloo")
+    old = _sys.stdout
+
+    def run(args):
+        buf = io.StringIO()
+        _sys.stdout = buf
+        try:
+            code = cli.main(list(args))
+        finally:
+            _sys.stdout = old
+        return code, buf.getvalue()
+
+    outs = []
+
+    def emit(args):
+        outs.append(run(args))
+
+    # phase 1: exhaustive metric, distributed build, 13 queries
+    emit(["build", "input"])
+    if rank == 0:
+        files = []
+        for root, _dirs, names in os.walk(idx_root):
+            for n in names:
+                if n.endswith(".sqlite"):
+                    files.append(os.path.relpath(
+                        os.path.join(root, n), idx_root))
+        outs.append((0, "".join(f + "\n" for f in sorted(files))))
+    for args in arglists[:13]:
+        emit(list(args) + ["input"])
+
+    # phase 2: filtered index (reference: metric-remove + filtered
+    # metric-add + rebuild), then the filtered query
+    dist.barrier()
+    if rank == 0:
+        emit(["metric-remove", "input", "mymet"])
+        emit(["metric-add", "input", "-f",
+              '{ "eq": [ "req.method", "GET" ] }', "-b",
+              "timestamp[date,field=time,aggr=lquantize,step=86400]",
+              "mymet"])
+    dist.barrier()
+    emit(["build", "input"])
+    emit(list(arglists[13]) + ["input"])
+
+    # phase 3: datasource filter always applied (mrm -r; update
+    # datasource filter; add bycode; rebuild; two queries)
+    dist.barrier()
+    if rank == 0:
+        shutil.rmtree(idx_root)
+        emit(["datasource-update", "input",
+              '--filter={ "eq": [ "req.method", "GET" ] }'])
+        emit(["metric-add", "input", "bycode", "-b",
+              "res.statusCode"])
+    dist.barrier()
+    emit(["build", "input"])
+    for args in arglists[14:16]:
+        emit(list(args) + ["input"])
+
+    out_q.put((rank, outs))
+    dist.barrier()
+    import torch.distributed as tdist
+    tdist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_index_manta_goldens_sharded(tmp_path):
+    """tst.index_manta.sh.out replayed through the 2-rank sharded
+    backend: partitioned distributed builds (the reference's Manta
+    map/reduce build), index file listing, the scan_testcases
+    answered from the index, a filtered index, and the
+    datasource-filter-always-applied rebuild — byte-for-byte
+    (reference tests/dn/manta/tst.index_manta.sh)."""
+    import torch.multiprocessing as mp
+    text = open(os.path.join(
+        REF, "dn", "manta", "tst.index_manta.sh.out")).read()
+    parts = re.split(r"^# dn (.*)$", text, flags=re.M)
+    preamble = parts[0]
+    sections = [(parts[i],
+                 parts[i + 1][1:] if parts[i + 1].startswith("\n")
+                 else parts[i + 1])
+                for i in range(1, len(parts), 2)]
+    assert len(sections) == 16
+
+    idx_root = str(tmp_path / "idx")
+    cfgfile = str(tmp_path / "rc.json")
+    from dragnet_amd import config as mod_config
+    cfg = mod_config.DragnetConfig()
+    cfg.datasource_add(mod_config.Datasource(
+        name="input", backend="sharded", path=DATA,
+        time_format="%Y/%m-%d", time_field="time",
+        index_path=idx_root))
+    cfg.metric_add(mod_config.Metric(
+        name="mymet", datasource="input", breakdowns=[
+            {"name": "timestamp", "date": "", "field": "time",
+             "aggr": "lquantize", "step": 86400},
+            {"name": "host"}, {"name": "operation"},
+            {"name": "req.caller"}, {"name": "req.method"},
+            {"name": "latency", "aggr": "quantize"}]))
+    mod_config.save_config(cfg, cfgfile)
+
+    arglists = [["query"] + split_args(a)[1:] for a, _ in sections]
+    ctx = mp.get_context("spawn")
+    out_q = ctx.Queue()
+    procs = [ctx.Process(
+        target=_worker_index_manta,
+        args=(r, 2, 29549, cfgfile, idx_root, arglists, out_q))
+        for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, outs = out_q.get(timeout=240)
+        results[rank] = outs
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    r0 = results[0]
+    assert all(code == 0 for code, _ in r0)
+    # rank 0's stream: build(no output) + listing + 13 queries +
+    # [2 config cmds] + build + query13 + [2 config cmds] + build +
+    # queries 14,15
+    texts = [out for _code, out in r0]
+    assert texts[0] == ""                       # build prints nothing
+    assert texts[1] == preamble                 # the mfind listing
+    for i in range(13):
+        assert norm(texts[2 + i]) == expected_body(sections[i][1]), \
+            "# dn " + sections[i][0]
+    # config mutations print nothing
+    assert texts[15] == texts[16] == ""
+    assert texts[17] == ""                      # build 2
+    assert norm(texts[18]) == expected_body(sections[13][1])
+    assert texts[19] == texts[20] == ""
+    assert texts[21] == ""                      # build 3
+    assert norm(texts[22]) == expected_body(sections[14][1])
+    assert norm(texts[23]) == expected_body(sections[15][1])
+    # rank 1 prints nothing anywhere
+    assert all(code == 0 and out == "" for code, out in results[1])

@@ -239,3 +239,53 @@ def test_serializer_reducer_roundtrip():
     assert punted == []
     assert aggs[0].table == src.table
     assert sum(aggs[0].table.values()) == sum(src.table.values())
+
+
+from hypothesis import given, settings  # noqa: E402
+from hypothesis import strategies as st  # noqa: E402
+
+_scalar = st.one_of(
+    st.none(), st.booleans(),
+    st.integers(min_value=-2**70, max_value=2**70),
+    st.floats(allow_nan=False, allow_infinity=False),
+    st.text(max_size=10),
+    st.lists(st.integers(min_value=0, max_value=3), max_size=2),
+)
+_fieldname = st.sampled_from(
+    ["a", "t", "lat", "ts", "req.method", "res.statusCode",
+     "__dn_metric", "junk", "ünïcode"])
+_point_line = st.builds(
+    lambda f, v: json.dumps({"fields": f, "value": v}).encode(),
+    st.dictionaries(_fieldname, _scalar, max_size=6),
+    st.one_of(st.integers(min_value=-10**6, max_value=10**6),
+              st.sampled_from([0.5, 0.25, 1.75, 3.0])))
+_raw_line = st.binary(max_size=24).map(lambda b: b.replace(b"\n", b"x"))
+
+
+@settings(max_examples=150, deadline=None)
+@given(lines=st.lists(st.one_of(_point_line, _raw_line, st.just(b"")),
+                      max_size=25))
+def test_property_reducer_differential(lines):
+    """Any mix of point-shaped and arbitrary lines reduces
+    identically through the C++ fast path + punts and the pure
+    Python loop (values restricted to exactly-representable sums so
+    addition order cannot matter)."""
+    # lines whose json is a non-dict would crash BOTH paths
+    # identically (AttributeError in the CLI); keep the differential
+    # on the non-crashing subset
+    keep = []
+    for ln in lines:
+        try:
+            v = json.loads(ln.strip() or b"{}")
+            if not isinstance(v, dict):
+                continue
+            if not isinstance(v.get("fields", {}), dict):
+                continue
+            mi = v.get("fields", {}).get("__dn_metric")
+            if isinstance(mi, int) and 0 <= mi < len(QUERIES) \
+                    and "value" not in v:
+                continue  # KeyError in both paths
+        except ValueError:
+            pass
+        keep.append(ln)
+    check(keep) if keep else None

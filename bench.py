@@ -57,9 +57,10 @@ def main():
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--mb", type=int,
-                    default=int(os.environ.get("DRAGNET_BENCH_MB", 512)),
-                    help="per-GPU NDJSON pool size (MB; 512 measured "
-                         "+1.3%% over 256 — fewer pass boundaries)")
+                    default=int(os.environ.get("DRAGNET_BENCH_MB", 1024)),
+                    help="per-GPU NDJSON pool size (MB; swept 256->512"
+                         "->1024->2048: 1024 is +2.6%% over 512, 2048 "
+                         "flat — fewer pass boundaries)")
     ap.add_argument("--device-resident", action="store_true",
                     help="skip per-step H2D (data already in HBM): "
                          "measures kernel-side scan throughput")

@@ -64,6 +64,14 @@ def main(argv=None):
         elif argv[0] == "--dry-run":
             argv.pop(0)
             dry_run = True
+        elif argv[0] == "--help":
+            sys.stdout.write(
+                "usage: python -m dragnet_amd.launch [--gpus N] "
+                "[--dry-run] SUBCOMMAND ...\n"
+                "Runs `dn SUBCOMMAND ...` as N ranks (one per GPU) "
+                "via torch.distributed.run;\n--dry-run prints the "
+                "job definition as JSON.\n")
+            return 0
         else:
             sys.stderr.write("dn-launch: unknown option %s\n" % argv[0])
             return 2

@@ -54,3 +54,30 @@ def test_count_conservation(fixture_tree):
     parser = dict(res.stages)["json parser"]
     assert parser["ninputs"] == 2254
     assert parser["noutputs"] + parser["invalid json"] == 2254
+
+
+def test_examples_build_and_query(dn, fixture_tree, tmp_path):
+    """The committed examples/ configs are valid and drive a real
+    build + query end-to-end (BASELINE config #4 uses
+    examples/index-muskie-local.json)."""
+    import json
+    import os
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    exdir = os.path.join(root, "examples")
+    with open(os.path.join(exdir, "index-muskie-local.json")) as f:
+        excfg = json.load(f)
+    assert excfg["metrics"][0]["breakdowns"]
+    with open(os.path.join(exdir, "query-muskie-requests.json")) as f:
+        json.load(f)
+
+    idx = str(tmp_path / "exidx")
+    r = dn("datasource-add", "exsrc", "--path=" + fixture_tree,
+           "--time-field=time", "--time-format=%Y/%m-%d",
+           "--index-path=" + idx)
+    assert r.code == 0, r.err
+    r = dn("build", "--index-config=" +
+           os.path.join(exdir, "index-muskie-local.json"), "exsrc")
+    assert r.code == 0, r.err
+    r = dn("query", "-b", "req.method,res.statusCode", "exsrc")
+    assert r.code == 0, r.err
+    assert "VALUE" in r.out and len(r.out.splitlines()) > 2

@@ -549,3 +549,35 @@ def test_xpose_device_vs_host_staging(engines, tmp_path):
             os.environ.pop("DRAGNET_XPOSE_HOST", None)
     assert results["device"][0] == expected
     assert results["device"] == results["host"]
+
+
+@pytest.mark.gpu
+def test_dense_slots_knob(engines, tmp_path):
+    """DRAGNET_DENSE_SLOTS (the documented high-cardinality knob from
+    the r2 slot sweep) must not change results — 16384- and
+    32768-slot directories equal the default and the oracle."""
+    import json as _json
+    import random
+
+    from dragnet_amd.engine.gpu import GpuEngine
+    from dragnet_amd.query import query_load
+    cpu, _ = engines
+
+    rng = random.Random(11)
+    lines = []
+    for i in range(60000):
+        rec = {"a": "k%d" % rng.randint(0, 200),
+               "b": rng.randint(0, 30)}
+        lines.append(_json.dumps(rec))
+    f = tmp_path / "slots.log"
+    f.write_bytes(("\n".join(lines) + "\n").encode())
+
+    q = query_load(breakdown_specs="a,b")
+    c = cpu.scan([str(f)], [q])
+    for slots in ("16384", "32768"):
+        os.environ["DRAGNET_DENSE_SLOTS"] = slots
+        try:
+            g = GpuEngine().scan([str(f)], [q])
+        finally:
+            os.environ.pop("DRAGNET_DENSE_SLOTS", None)
+        assert_same(c, g)

@@ -246,3 +246,33 @@ def test_cli_stdin_scan_gpu(tmp_path):
         out[engine] = r.stdout
     assert out["gpu"] == out["cpu"]
     assert b"GET" in out["gpu"]
+
+
+def test_cli_index_pipe_gpu(dn, fixture_tree, tmp_path, monkeypatch):
+    """index-scan | index-read with the GPU engine on the map side ==
+    the CPU pipe (map scans run the fused kernel; the reduce runs the
+    native point codec)."""
+    one = os.path.join(fixture_tree, "2014", "05-01", "one.log")
+    outs = {}
+    for eng in ("cpu", "gpu"):
+        monkeypatch.setenv("DRAGNET_ENGINE", eng)
+        src = "psrc_" + eng
+        dst = "pdst_" + eng
+        assert dn("datasource-add", src, "--path=" + one,
+                  "--index-path=" + str(tmp_path / (eng + "0")),
+                  "--time-field=time").code == 0
+        assert dn("metric-add", src, "m", "-b",
+                  "operation,req.method").code == 0
+        r = dn("index-scan", src)
+        assert r.code == 0, r.err
+        assert dn("datasource-add", dst, "--path=/dev/null",
+                  "--index-path=" + str(tmp_path / ("idx_" + eng)),
+                  "--time-field=time").code == 0
+        assert dn("metric-add", dst, "m", "-b",
+                  "operation,req.method").code == 0
+        assert dn("index-read", dst, stdin=r.out.encode()).code == 0
+        q = dn("query", "-b", "operation", dst)
+        assert q.code == 0, q.err
+        outs[eng] = (r.out, q.out)
+    assert outs["gpu"] == outs["cpu"]
+    assert outs["gpu"][1] != ""

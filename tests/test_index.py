@@ -175,3 +175,43 @@ def test_index_scan_index_read_pipeline(dn, fixture_tree, tmp_path):
     qb = dn("query", "-b", "operation", "b")
     assert qa.out == qb.out
     assert qa.out != ""
+
+
+def test_index_read_codec_vs_python_cli(dn, fixture_tree, tmp_path,
+                                        monkeypatch):
+    """The native point codec and the pure-Python loop produce
+    identical indexes through the real `dn index-scan | dn
+    index-read` pipe (DRAGNET_PY_POINTS=1 forces Python end-to-end:
+    both the reduce parse and the emit serializer)."""
+    one = os.path.join(fixture_tree, "2014", "05-01", "one.log")
+    r = dn("datasource-add", "src", "--path=" + one,
+           "--index-path=" + str(tmp_path / "i0"),
+           "--time-field=time")
+    assert r.code == 0, r.err
+    r = dn("metric-add", "src", "m", "-b",
+           "operation,req.method,latency[aggr=quantize]")
+    assert r.code == 0, r.err
+
+    outs = {}
+    for mode in ("native", "python"):
+        if mode == "python":
+            monkeypatch.setenv("DRAGNET_PY_POINTS", "1")
+        else:
+            monkeypatch.delenv("DRAGNET_PY_POINTS", raising=False)
+        r = dn("index-scan", "src")
+        assert r.code == 0, r.err
+        points = r.out
+        name = "dst_" + mode
+        idx = str(tmp_path / ("idx_" + mode))
+        assert dn("datasource-add", name, "--path=/dev/null",
+                  "--index-path=" + idx,
+                  "--time-field=time").code == 0
+        assert dn("metric-add", name, "m", "-b",
+                  "operation,req.method,latency[aggr=quantize]"
+                  ).code == 0
+        assert dn("index-read", name, stdin=points.encode()).code == 0
+        q = dn("query", "-b", "operation,latency[aggr=quantize]", name)
+        assert q.code == 0, q.err
+        outs[mode] = (points, q.out)
+    assert outs["native"] == outs["python"]
+    assert outs["native"][1] != ""

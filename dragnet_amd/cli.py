@@ -176,7 +176,8 @@ _OPTDEFS = {
 }
 
 _SHORT = {"-f": "filter", "-b": "breakdowns", "-v": "verbose",
-          "-I": "interval"}
+          "-i": "interval", "-A": "after", "-B": "before",
+          "-n": "dry-run"}
 
 
 def parse_args(argv, allowed):

@@ -81,3 +81,22 @@ def test_examples_build_and_query(dn, fixture_tree, tmp_path):
     r = dn("query", "-b", "req.method,res.statusCode", "exsrc")
     assert r.code == 0, r.err
     assert "VALUE" in r.out and len(r.out.splitlines()) > 2
+
+
+def test_short_option_aliases(dn, fixture_tree):
+    """Reference short aliases -A/-B/-n/-i/-f/-b (bin/dn:146-215)."""
+    one = os.path.join(fixture_tree, "2014", "05-01", "one.log")
+    assert dn("datasource-add", "al", "--path=" + fixture_tree,
+              "--time-field=time",
+              "--time-format=%Y/%m-%d").code == 0
+    long = dn("scan", "--filter", '{"eq":["req.method","GET"]}',
+              "--breakdowns", "operation",
+              "--after", "2014-05-01", "--before", "2014-05-02", "al")
+    short = dn("scan", "-f", '{"eq":["req.method","GET"]}',
+               "-b", "operation",
+               "-A", "2014-05-01", "-B", "2014-05-02", "al")
+    assert long.code == 0 and short.code == 0, short.err
+    assert short.out == long.out and long.out != ""
+    dry = dn("scan", "-n", "al")
+    assert dry.code == 0
+    assert "would scan" in dry.err + dry.out

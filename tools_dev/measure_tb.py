@@ -1,8 +1,15 @@
 """TB-class single-rank file-scan leg (BASELINE config #5's scale
 label, single-GPU slice): build the largest corpus /dev/shm safely
 holds (capped at 1 TB), scan it engine-level with the flagship query,
-report GB/s.  Refuses to run without a ~192 GB free-RAM margin and
-removes the corpus afterwards."""
+report GB/s.
+
+DO NOT run this on a shared or RAM-constrained box.  The first
+attempt used tmpfs free space as the safety margin and took the box
+down: tmpfs `disk_usage().free` reflects the MOUNT size, not actual
+available RAM, so filling it can OOM the host.  The guard below now
+keys off /proc/meminfo MemAvailable with a 256 GB floor and a 60%
+cap, and the corpus is removed afterwards — but treat this tool as
+operator-supervised, not CI."""
 import json
 import multiprocessing as mp
 import os
@@ -26,15 +33,28 @@ def _copy_batch(jobs):
     return len(jobs)
 
 
+def mem_available_bytes():
+    with open("/proc/meminfo") as f:
+        for ln in f:
+            if ln.startswith("MemAvailable:"):
+                return int(ln.split()[1]) * 1024
+    return 0
+
+
 def main():
     shm = shutil.disk_usage("/dev/shm")
-    log("/dev/shm free: %.0f GB of %.0f" % (shm.free / GB,
-                                            shm.total / GB))
+    avail = mem_available_bytes()
+    log("/dev/shm free: %.0f GB of %.0f; MemAvailable: %.0f GB"
+        % (shm.free / GB, shm.total / GB, avail / GB))
+    # tmpfs free space is NOT available RAM — key the budget off
+    # MemAvailable (what the kernel can actually give us)
     target = min(shm.free - 192 * GB,
+                 int(avail * 0.6),
+                 avail - 256 * GB,
                  int(float(os.environ.get("TB_TARGET_GB", 1024)) * GB))
     if target < 300 * GB:
-        log("not enough /dev/shm headroom for a TB-class corpus; "
-            "refusing (need >= ~500 GB free)")
+        log("not enough ACTUAL free RAM for a TB-class corpus; "
+            "refusing (need >= ~560 GB MemAvailable)")
         return 1
     shutil.rmtree(ROOT, ignore_errors=True)
     os.makedirs(ROOT)

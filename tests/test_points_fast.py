@@ -20,6 +20,13 @@ from dragnet_amd.query import query_load
 pytest.importorskip("dragnet_amd.index._points")
 
 
+@pytest.fixture(autouse=True)
+def _force_native(monkeypatch):
+    """These tests exercise the codec itself — clear the env gate
+    that forces the pure-Python path."""
+    monkeypatch.delenv("DRAGNET_PY_POINTS", raising=False)
+
+
 QUERIES = [
     query_load(breakdown_specs="a,t[date,field=time],"
                                "lat[aggr=quantize],"

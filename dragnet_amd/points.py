@@ -65,8 +65,14 @@ def canonical(val, has_date=False):
     if isinstance(val, bool):
         return "true" if val else "false"
     if has_date:
-        # synthetic date fields carry unix seconds; keep numeric
-        return int(val) if isinstance(val, (int, float)) else val
+        # synthetic date fields carry unix seconds; keep numeric.
+        # Non-numeric values (possible only on the re-aggregation
+        # path — scan-side synthetics drop them first) coerce like
+        # any JS object key below, mirroring skinner: a raw
+        # passthrough here crashed on list values (found by the
+        # 300-seed codec soak).
+        if isinstance(val, (int, float)):
+            return int(val)
     if isinstance(val, (int, float)):
         return js_num_str(val)
     if isinstance(val, str):

@@ -170,3 +170,27 @@ def test_query_validation():
         QueryConfig(time_after="2014-01-02", time_before="2014-01-01")
     q = query_load(breakdown_specs="ts[field=time,date]")
     assert q.synthetic == [{"name": "ts", "field": "time", "date": ""}]
+
+
+def test_date_column_nonscalar_reaggregation():
+    """A list/dict value reaching a DATE column on the
+    re-aggregation path (scan-side synthetics can't filter it there)
+    coerces like any JS object key instead of crashing — found by
+    the codec soak (points.canonical raw-passthrough bug)."""
+    from dragnet_amd.points import Aggregator
+    from dragnet_amd.query import query_load
+    q = query_load(
+        breakdown_specs="t[date,field=time,aggr=lquantize,step=60],a")
+    agg = Aggregator(q)
+    # lquantize'd date column: non-numeric -> nonnumeric drop
+    assert agg.write({"fields": {"t": [1], "a": "x"}, "value": 1}) \
+        is False
+    assert agg.ndropped_nonnumeric == 1
+    # plain (non-aggregated) date column: JS string coercion
+    q2 = query_load(breakdown_specs="t[date,field=time],a")
+    agg2 = Aggregator(q2)
+    assert agg2.write({"fields": {"t": [1, None], "a": "x"},
+                       "value": 2}) is True
+    assert agg2.write({"fields": {"t": {"z": 1}, "a": "x"},
+                       "value": 3}) is True
+    assert agg2.table == {("1,", "x"): 2, ("[object Object]", "x"): 3}

@@ -130,11 +130,16 @@ def _dispatch(argv):
     except UsageError as e:
         return usage(str(e))
     except (FatalError, QueryError, mod_config.ConfigError,
-            krill.KrillError) as e:
+            krill.KrillError, _sink_error()) as e:
         sys.stderr.write("dn: %s\n" % e)
         return 1
     except BrokenPipeError:
         return 0
+
+
+def _sink_error():
+    from .index.sink import SinkError
+    return SinkError
 
 
 def usage(msg, full=False):

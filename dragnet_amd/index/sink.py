@@ -27,6 +27,13 @@ import os
 INDEX_VERSION = "2.0.0"
 
 
+class SinkError(Exception):
+    """Index materialization failed (e.g. a breakdown name that
+    collides after [.-]->_ escaping, or an SQL-keyword column — the
+    reference's unquoted CREATE TABLE fails identically; we surface
+    it as a clean `dn:` error instead of a traceback)."""
+
+
 def sqlite3_escape(name):
     """Column-name escaping: [.-] -> _ (reference lib/index-sink.js:232)."""
     return name.replace(".", "_").replace("-", "_")
@@ -113,6 +120,20 @@ class IndexSink(object):
             for i, m in enumerate(self.metrics)]
 
     def _init_db(self):
+        try:
+            self._init_db_inner()
+        except Exception as e:
+            msg = str(e)
+            try:
+                self.abort()
+            except Exception:
+                pass
+            if "sqlite" in msg or "column" in msg or "syntax" in msg \
+                    or type(e).__module__ == "sqlite3":
+                raise SinkError("cannot materialize index: %s" % msg)
+            raise
+
+    def _init_db_inner(self):
         creates, inserts = self._schema_sql()
         if self._cs is not None:
             for sql in creates:

@@ -112,3 +112,25 @@ def test_native_vs_python_sink_identical(tmp_path):
         return out
 
     assert dump(f_native) == dump(f_py)
+
+
+def test_colliding_column_names_clean_error(dn, fixture_tree,
+                                            tmp_path):
+    """Breakdown names that collide after [.-]->_ escaping (or hit an
+    SQL keyword) fail the build with a clean `dn:` error, matching
+    the reference's unquoted CREATE TABLE failure mode — never a
+    traceback, and no stray tmp file."""
+    one = os.path.join(fixture_tree, "2014", "05-01", "one.log")
+    idx = str(tmp_path / "cidx")
+    assert dn("datasource-add", "c", "--path=" + one,
+              "--index-path=" + idx, "--time-field=time").code == 0
+    assert dn("metric-add", "c", "m", "-b", "a-b,a.b").code == 0
+    r = dn("build", "c")
+    assert r.code == 1
+    assert r.err.startswith("dn: cannot materialize index:")
+    assert "duplicate column" in r.err
+    # tmp files cleaned up (abort path)
+    leftovers = []
+    for root, _d, names in os.walk(idx):
+        leftovers += [n for n in names if not n.endswith(".sqlite")]
+    assert leftovers == [], leftovers

@@ -108,8 +108,19 @@ def main():
         import torch.distributed as torch_dist
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29601")
-        torch_dist.init_process_group(backend, rank=rank,
-                                      world_size=world)
+        # keep stdout pure for the one-JSON-line contract: the gloo /
+        # RCCL init banners printf to fd 1, so route fd 1 to stderr
+        # around init
+        sys.stdout.flush()
+        saved_fd1 = os.dup(1)
+        os.dup2(2, 1)
+        try:
+            torch_dist.init_process_group(backend, rank=rank,
+                                          world_size=world)
+        finally:
+            sys.stdout.flush()
+            os.dup2(saved_fd1, 1)
+            os.close(saved_fd1)
         dist = torch_dist
 
     from dragnet_amd.distributed import merge_tables_tensor

@@ -9,6 +9,8 @@ Sections:
   2. pipeline conservation — Σ drops + outputs == inputs at every stage
   3. merge wire format   — encode/rebuild round-trip of the RCCL merge
   4. CLI argv fuzz       — random argv must error cleanly, never raise
+  5. skinner conservation — the weighted json-skinner pipeline
+  6. config CRUD fuzz    — registry stays valid/atomic under random ops
 """
 import io
 import json
@@ -204,11 +206,126 @@ def soak_argv():
     return bad
 
 
+def soak_skinner():
+    from dragnet_amd.query import query_load
+    from dragnet_amd.scan_cpu import ScanPipeline
+    DROPS = ("invalid json", "nfilteredout", "nfailedeval", "undef",
+             "baddate", "nonnumeric")
+    QS = [{}, {"breakdown_specs": "a"},
+          {"breakdown_specs": "lat[aggr=quantize]"},
+          {"filter": {"eq": ["a", "x"]}, "breakdown_specs": "a,b"}]
+    bad = 0
+    for seed in range(n(200)):
+        rng = random.Random(seed)
+        lines = []
+        for _ in range(50):
+            c = rng.random()
+            if c < 0.6:
+                p = {"fields": {k: rng.choice(
+                        ["x", "y", 3, None, "26", [1]])
+                        for k in ("a", "b", "lat")
+                        if rng.random() < 0.7},
+                     "value": rng.choice([1, 2, 0.5, -1, 10**9])}
+                lines.append(json.dumps(p).encode())
+            elif c < 0.8:
+                lines.append(json.dumps(
+                    {"notfields": 1, "value": "x"}).encode())
+            else:
+                lines.append(rng.choice([b"", b"junk {", b"[1]"]))
+        data = b"".join(ln + b"\n" for ln in lines)
+        for qkw in QS:
+            p = ScanPipeline(query_load(**qkw),
+                             data_format="json-skinner")
+            p.write_bytes(data)
+            p.finish()
+            stages = p.counter_stages()
+            prev = None
+            for name, cnt in stages:
+                if name == "Aggregator":
+                    continue
+                drops = sum(cnt.get(k, 0) for k in DROPS)
+                if cnt["ninputs"] != cnt["noutputs"] + drops:
+                    bad += 1
+                    print("skinner SEED %d %r %s" % (seed, qkw, name))
+                prev = cnt["noutputs"]
+            if dict(stages)["Aggregator"]["ninputs"] != prev:
+                bad += 1
+                print("skinner SEED %d %r chain" % (seed, qkw))
+    return bad
+
+
+def soak_config():
+    bad = 0
+    for seed in range(n(150)):
+        cfgfile = tempfile.mktemp()
+        os.environ["DRAGNET_CONFIG"] = cfgfile
+        os.environ["DRAGNET_ENGINE"] = "cpu"
+        from dragnet_amd import cli
+        rng = random.Random(seed)
+        names = ["a", "b", "wëird", "x-y", "s" * 30]
+        live = set()
+        for step in range(25):
+            op = rng.random()
+            name = rng.choice(names)
+            o, e = sys.stdout, sys.stderr
+            sys.stdout, sys.stderr = io.StringIO(), io.StringIO()
+            try:
+                if op < 0.35:
+                    if cli.main(["datasource-add", name,
+                                 "--path=/tmp/x%d"
+                                 % rng.randint(0, 3)]) == 0:
+                        live.add(name)
+                elif op < 0.5:
+                    if cli.main(["datasource-remove", name]) == 0:
+                        live.discard(name)
+                elif op < 0.65:
+                    cli.main(["datasource-update", name,
+                              "--time-field=t%d" % rng.randint(0, 2)])
+                elif op < 0.8:
+                    cli.main(["metric-add", name,
+                              "m%d" % rng.randint(0, 2), "-b",
+                              rng.choice(["a", "lat[aggr=quantize]",
+                                          "bad["])])
+                else:
+                    cli.main(["metric-remove", name,
+                              "m%d" % rng.randint(0, 2)])
+            except Exception as ex:
+                bad += 1
+                e.write("config SEED %d: %r\n" % (seed, ex))
+            finally:
+                sys.stdout, sys.stderr = o, e
+            if os.path.exists(cfgfile):
+                try:
+                    json.load(open(cfgfile))
+                except ValueError:
+                    bad += 1
+                    print("config SEED %d: corrupt file" % seed)
+            out = io.StringIO()
+            o = sys.stdout
+            sys.stdout = out
+            try:
+                assert cli.main(["datasource-list"]) == 0
+            finally:
+                sys.stdout = o
+            listed = {ln.split()[0]
+                      for ln in out.getvalue().splitlines()
+                      if ln.strip() and not ln.startswith("DATASOURCE")}
+            if listed != live:
+                bad += 1
+                print("config SEED %d: list mismatch" % seed)
+                break
+        if os.path.exists(cfgfile):
+            os.unlink(cfgfile)
+    return bad
+
+
 def main():
     total = 0
     for name, fn in (("codec", soak_codec),
                      ("pipeline", soak_pipeline),
-                     ("wire", soak_wire), ("argv", soak_argv)):
+                     ("wire", soak_wire), ("argv", soak_argv),
+                     ("skinner", soak_skinner),
+                     ("config", soak_config)):
         bad = fn()
         print("%s: %s" % (name, "CLEAN" if bad == 0
                           else "%d FAILURES" % bad))

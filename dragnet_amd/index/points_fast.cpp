@@ -590,6 +590,7 @@ py::tuple reduce_tagged(
 void esc_json(std::string& out, PyObject* u) {
   // matches c_encode_basestring_ascii: ensure_ascii, lowercase hex
   static const char* hexd = "0123456789abcdef";
+  if (PyUnicode_READY(u) != 0) throw py::error_already_set();
   Py_ssize_t n = PyUnicode_GET_LENGTH(u);
   int kind = PyUnicode_KIND(u);
   const void* dat = PyUnicode_DATA(u);
@@ -646,6 +647,10 @@ bool emit_scalar(std::string& out, PyObject* o) {
     if (r == nullptr) throw py::error_already_set();
     Py_ssize_t len;
     const char* c = PyUnicode_AsUTF8AndSize(r, &len);
+    if (c == nullptr) {
+      Py_DECREF(r);
+      throw py::error_already_set();
+    }
     out.append(c, static_cast<size_t>(len));
     Py_DECREF(r);
     return true;
@@ -661,6 +666,10 @@ bool emit_scalar(std::string& out, PyObject* o) {
     if (r == nullptr) throw py::error_already_set();
     Py_ssize_t len;
     const char* c = PyUnicode_AsUTF8AndSize(r, &len);
+    if (c == nullptr) {
+      Py_DECREF(r);
+      throw py::error_already_set();
+    }
     out.append(c, static_cast<size_t>(len));
     Py_DECREF(r);
     return true;

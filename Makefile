@@ -1,17 +1,28 @@
-# dragnet_amd build/test entry points (the reference's make test analog)
-.PHONY: build test test-gpu bench lint
+# dragnet_amd developer targets (the reference's Makefile analog:
+# `make test` there runs catest -a; here the pytest suites).
+
+PY ?= python3
+
+.PHONY: all build test test-gpu soak bench clean
+
+all: build
 
 build:
-	PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+	PYTORCH_ROCM_ARCH=gfx950 $(PY) setup.py build_ext --inplace
 
 test:
-	python -m pytest tests -q -m "not gpu"
+	$(PY) -m pytest tests -q -m "not gpu"
 
 test-gpu:
-	python -m pytest tests -q -m gpu
+	$(PY) -m pytest tests -q -m gpu
+
+soak:
+	$(PY) tools_dev/soak_cpu.py
 
 bench:
-	python bench.py --gpus 1 --steps 10 --warmup 3
+	$(PY) bench.py --steps 10 --warmup 3
 
-lint:
-	python -m pyflakes dragnet_amd bench.py __graft_entry__.py || true
+clean:
+	rm -rf build dragnet_amd/ops/_dragnet_hip*.so \
+	    dragnet_amd/index/_csink*.so dragnet_amd/index/_points*.so \
+	    dragnet_amd/ops/hip/*_hip.hip

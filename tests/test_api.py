@@ -100,3 +100,21 @@ def test_short_option_aliases(dn, fixture_tree):
     dry = dn("scan", "-n", "al")
     assert dry.code == 0
     assert "would scan" in dry.err + dry.out
+
+
+def test_api_index_config(fixture_tree, tmp_path, monkeypatch):
+    """api.index_config returns the metric set the CLI's
+    `dn index-config` prints."""
+    monkeypatch.setenv("DRAGNET_CONFIG", str(tmp_path / "rc.json"))
+    monkeypatch.setenv("DRAGNET_ENGINE", "cpu")
+    from dragnet_amd import api, config as mod_config
+    cfg = mod_config.DragnetConfig()
+    cfg.datasource_add(mod_config.Datasource(
+        name="s", backend="file", path=fixture_tree))
+    cfg.metric_add(mod_config.Metric(
+        name="m", datasource="s",
+        breakdowns=[{"name": "operation"}]))
+    mod_config.save_config(cfg, str(tmp_path / "rc.json"))
+    out = api.index_config("s")
+    assert out["metrics"][0]["name"] == "m"
+    assert out["metrics"][0]["breakdowns"][0]["name"] == "operation"

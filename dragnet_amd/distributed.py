@@ -119,7 +119,16 @@ def _encode_table(agg, nk):
                 codes[i, j] = np.float64(el).view(np.int64)
                 tp |= _EL_F64 << (2 * j)
             else:
-                codes[i, j] = int(el)
+                iv = int(el)
+                if -(1 << 63) <= iv < (1 << 63):
+                    codes[i, j] = iv
+                else:
+                    # lquantize ordinals of absurd magnitudes exceed
+                    # int64: ship as f64 bits (JS-float key semantics
+                    # — the reference's ordinals are floats there
+                    # anyway); every rank rebuilds the same float key
+                    codes[i, j] = np.float64(float(iv)).view(np.int64)
+                    tp |= _EL_F64 << (2 * j)
         tagpack[i] = tp
     return codes, tagpack, vals, strings
 

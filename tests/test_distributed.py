@@ -429,3 +429,24 @@ def test_merge_float_weights_and_negative_ordinals():
         assert table == expected
         assert isinstance(table[("int", 1)], int)
         assert isinstance(table[("r0", -2)], float)
+
+
+def test_wire_encode_huge_lquantize_ordinal():
+    """lquantize ordinals beyond int64 (floor(1e300/step) is a
+    ~1000-bit Python int) must survive the dense wire encode as f64
+    keys instead of raising OverflowError (single-process slice of
+    the C1 format)."""
+    from dragnet_amd.distributed import _encode_table, _rebuild_table
+    from dragnet_amd.points import Aggregator
+    from dragnet_amd.query import query_load
+    q = query_load(breakdown_specs="x[aggr=lquantize,step=7]")
+    a = Aggregator(q)
+    assert a.write({"fields": {"x": 1e300}, "value": 3})
+    assert a.write({"fields": {"x": -1e300}, "value": 4})
+    assert a.write({"fields": {"x": 2}, "value": 5})
+    codes, tags, vals, strings = _encode_table(a, 1)
+    out = _rebuild_table(q, codes, tags, vals, strings)
+    assert sum(out.table.values()) == 12
+    assert (0,) in out.table and out.table[(0,)] == 5
+    huge = [k for k, in out.table if isinstance(k, float)]
+    assert len(huge) == 2 and huge[0] != huge[1]

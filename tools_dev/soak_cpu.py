@@ -11,6 +11,7 @@ Sections:
   4. CLI argv fuzz       — random argv must error cleanly, never raise
   5. skinner conservation — the weighted json-skinner pipeline
   6. config CRUD fuzz    — registry stays valid/atomic under random ops
+  7. index round-trip    — write_index -> IndexQuerier == direct table
 """
 import io
 import json
@@ -319,13 +320,69 @@ def soak_config():
     return bad
 
 
+def soak_index():
+    from dragnet_amd.datasource.file import write_index
+    from dragnet_amd.index.query import IndexQuerier
+    from dragnet_amd.points import Aggregator
+    from dragnet_amd.query import query_load
+    bad = 0
+    for seed in range(n(150)):
+        rng = random.Random(seed)
+        tmp = tempfile.mkdtemp()
+        nbd = rng.randint(1, 4)
+        bds = []
+        used = set()
+        for i in range(nbd):
+            name = rng.choice(["a", "b", "c", "lat", "t.s", "x-y"])
+            if name in used:
+                name += str(i)
+            used.add(name)
+            kind = rng.random()
+            if kind < 0.3:
+                bds.append({"name": name, "aggr": "quantize"})
+            elif kind < 0.5:
+                bds.append({"name": name, "aggr": "lquantize",
+                            "step": rng.choice([7, 60, 3600])})
+            else:
+                bds.append({"name": name})
+        metrics = [{"name": "m",
+                    "breakdowns": [dict(b) for b in bds]}]
+        q = query_load(breakdown_specs=",".join(
+            "%s[aggr=%s%s]" % (b["name"], b["aggr"],
+                               ",step=%d" % b["step"]
+                               if b["aggr"] == "lquantize" else "")
+            if "aggr" in b else b["name"] for b in bds))
+        src = Aggregator(q)
+        for _ in range(rng.randint(1, 120)):
+            fields = {}
+            for b in bds:
+                if "aggr" in b:
+                    fields[b["name"]] = rng.choice(
+                        [0, 1, 7, 26, 100, 4096, "26", 2.5])
+                else:
+                    fields[b["name"]] = rng.choice(
+                        ["x", "y", "", "z-9", None, 200, True])
+            src.write({"fields": dict(fields),
+                       "value": rng.randint(1, 9)})
+        pts = src.points()
+        for p in pts:
+            p["fields"]["__dn_metric"] = 0
+        write_index(tmp, metrics, "all", pts)
+        iq = IndexQuerier(os.path.join(tmp, "all"))
+        if iq.run(q).table != src.table:
+            bad += 1
+            print("index SEED %d mismatch" % seed)
+    return bad
+
+
 def main():
     total = 0
     for name, fn in (("codec", soak_codec),
                      ("pipeline", soak_pipeline),
                      ("wire", soak_wire), ("argv", soak_argv),
                      ("skinner", soak_skinner),
-                     ("config", soak_config)):
+                     ("config", soak_config),
+                     ("index", soak_index)):
         bad = fn()
         print("%s: %s" % (name, "CLEAN" if bad == 0
                           else "%d FAILURES" % bad))
